@@ -1,0 +1,157 @@
+"""Op dispatch layer: HIP/CDNA4 extension on GPU, eager reference on CPU.
+
+The hand-written gfx950 kernels live in ``csrc/`` and are built in-tree into
+``distegnn_amd/ops/_hip_ext*.so`` by ``python -m distegnn_amd.ops.build``
+(also driven by ``__graft_entry__.build()``). On a GPU box the HIP path is
+mandatory: calling one of these ops on a CUDA tensor without the extension
+raises, so a silent eager fallback can never masquerade as the native path.
+Set ``DISTEGNN_ALLOW_EAGER_GPU=1`` only for debugging.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import _hip_ext  # built in-tree; travels with the repo snapshot
+
+        _EXT = _hip_ext
+    except ImportError:
+        try:
+            import importlib
+
+            _EXT = importlib.import_module("_hip_ext")
+        except ImportError as e:  # pragma: no cover - error path
+            _EXT_ERR = str(e)
+            _EXT = None
+    return _EXT
+
+
+def hip_ext():
+    """The loaded HIP extension module, or None (CPU-only environment)."""
+    return _load_extension()
+
+
+def _require_ext(opname: str):
+    ext = _load_extension()
+    if ext is None:
+        if os.environ.get("DISTEGNN_ALLOW_EAGER_GPU") == "1":
+            return None
+        raise RuntimeError(
+            f"distegnn_amd op '{opname}' called on a CUDA tensor but the HIP "
+            f"extension is not built/importable ({_EXT_ERR}). Build it with "
+            f"`python -m distegnn_amd.ops.build` (gfx950). Refusing to fall "
+            f"back to eager on GPU; set DISTEGNN_ALLOW_EAGER_GPU=1 to debug."
+        )
+    return ext
+
+
+class _SegmentReduceFn(torch.autograd.Function):
+    """CSR segmented sum/mean over row-sorted edge data (HIP forward)."""
+
+    @staticmethod
+    def forward(ctx, data, row, rowptr, num_segments, mean):
+        ext = _require_ext("segment_reduce")
+        if ext is None:  # debug-only eager fallback
+            out = (reference.segment_mean if mean else reference.segment_sum)(
+                data, row, num_segments)
+        else:
+            out = ext.segment_reduce_csr(data, rowptr, bool(mean))
+        ctx.save_for_backward(row, rowptr)
+        ctx.mean = mean
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        row, rowptr = ctx.saved_tensors
+        gout = gout.contiguous()
+        if ctx.mean:
+            deg = (rowptr[1:] - rowptr[:-1]).clamp(min=1).to(gout.dtype)
+            gout = gout / deg.unsqueeze(-1)
+        gdata = gout.index_select(0, row)
+        return gdata, None, None, None, None
+
+
+def segment_sum(data: torch.Tensor, row: torch.Tensor, num_segments: int,
+                rowptr: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if data.is_cuda and rowptr is not None:
+        return _SegmentReduceFn.apply(data.contiguous(), row, rowptr,
+                                      num_segments, False)
+    return reference.segment_sum(data, row, num_segments)
+
+
+def segment_mean(data: torch.Tensor, row: torch.Tensor, num_segments: int,
+                 rowptr: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if data.is_cuda and rowptr is not None:
+        return _SegmentReduceFn.apply(data.contiguous(), row, rowptr,
+                                      num_segments, True)
+    return reference.segment_mean(data, row, num_segments)
+
+
+class _GraphPoolFn(torch.autograd.Function):
+    """Per-graph sum/mean pooling over ptr-delimited node blocks."""
+
+    @staticmethod
+    def forward(ctx, x, batch, ptr, num_graphs, mean):
+        ext = _require_ext("graph_pool")
+        if ext is None:
+            out = (reference.graph_mean_pool if mean else reference.graph_sum_pool)(
+                x, batch, num_graphs)
+        else:
+            out = ext.segment_reduce_csr(x, ptr, bool(mean))
+        ctx.save_for_backward(batch, ptr)
+        ctx.mean = mean
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        batch, ptr = ctx.saved_tensors
+        gout = gout.contiguous()
+        if ctx.mean:
+            cnt = (ptr[1:] - ptr[:-1]).clamp(min=1).to(gout.dtype)
+            gout = gout / cnt.view(-1, *([1] * (gout.dim() - 1)))
+        gx = gout.index_select(0, batch)
+        return gx, None, None, None, None
+
+
+def graph_sum_pool(x: torch.Tensor, batch: torch.Tensor, num_graphs: int,
+                   ptr: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if x.is_cuda and ptr is not None:
+        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs, False)
+    return reference.graph_sum_pool(x, batch, num_graphs)
+
+
+def graph_mean_pool(x: torch.Tensor, batch: torch.Tensor, num_graphs: int,
+                    ptr: Optional[torch.Tensor] = None,
+                    counts: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if x.is_cuda and ptr is not None:
+        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs, True)
+    return reference.graph_mean_pool(x, batch, num_graphs, counts=counts)
+
+
+def radius_graph(pos: torch.Tensor, r: float, loop: bool = False) -> torch.Tensor:
+    """Directed radius graph, row-sorted. GPU: HIP cell-list kernel."""
+    if pos.is_cuda and r is not None and r >= 0:
+        ext = _require_ext("radius_graph")
+        if ext is not None:
+            return ext.radius_graph(pos.contiguous().float(), float(r))
+    return reference.radius_graph(pos, r, loop=loop)
+
+
+__all__ = [
+    "segment_sum", "segment_mean", "graph_sum_pool", "graph_mean_pool",
+    "radius_graph", "hip_ext", "reference",
+]

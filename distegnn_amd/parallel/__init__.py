@@ -1,0 +1,20 @@
+from .comm import (
+    GradBucket,
+    all_reduce_sum_differentiable,
+    barrier,
+    check_model_parameters,
+    destroy,
+    fused_weighted_average_reduce,
+    global_counts,
+    init_distributed,
+    is_distributed,
+    rank,
+    world_size,
+)
+
+__all__ = [
+    "GradBucket", "all_reduce_sum_differentiable", "barrier",
+    "check_model_parameters", "destroy", "fused_weighted_average_reduce",
+    "global_counts", "init_distributed", "is_distributed", "rank",
+    "world_size",
+]

@@ -1,0 +1,294 @@
+"""Training runtime: epoch/step loops, loss assembly, checkpointing, logging.
+
+Behavioral parity with reference utils/train.py:17-289 (train_single_epoch /
+train): same loss weighting math (node-count weighted global MSE, sum
+semantics under data parallelism), same MMD formula, same gradient
+accumulation + clip-0.3 rule, same checkpoint dict layout and file names
+(best_model.pth / last_model.pth under {log_dir}/{exp_name}/state_dict,
+including the DDP 'module.' key prefix when world_size > 1), same JSON log
+(log.json: [best_log_dict, log_dict, config]), same early-stop broadcast.
+
+Deliberate MI355X-first deviations (same math, faster):
+* gradient sync is ONE flat all_reduce(SUM) per optimizer step
+  (parallel.GradBucket) instead of DDP's per-backward average + loss *
+  world_size; gradients entering clip/step are bit-identical in expectation.
+* the per-step logging all_reduce of the loss (train.py:109) is deferred:
+  local weighted losses accumulate on device and are reduced once per epoch.
+* the per-step loader-lockstep all_gather debug check (train.py:55-61) is
+  opt-in via ``debug_lockstep`` (config train.debug_lockstep or env
+  DISTEGNN_DEBUG_LOCKSTEP=1) rather than always-on in the hot loop.
+* MMD is vectorized on device (runtime/losses.py) — no per-graph host loop.
+"""
+
+from __future__ import annotations
+
+import contextlib
+import json
+import os
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+from ..parallel import comm
+from ..parallel.comm import GradBucket
+from .losses import mmd_loss
+
+try:
+    from tqdm import tqdm
+except ImportError:  # pragma: no cover
+    tqdm = None
+
+
+def state_dict_for_save(model: nn.Module, world_size: int) -> dict:
+    """Reference checkpoint key parity: DDP runs save 'module.'-prefixed keys
+    (reference saves model.state_dict() of the DDP wrapper, train.py:237)."""
+    sd = model.state_dict()
+    if world_size > 1:
+        sd = {f"module.{k}": v for k, v in sd.items()}
+    return sd
+
+
+def load_state_dict_compat(model: nn.Module, sd: dict):
+    """Accept both bare and 'module.'-prefixed checkpoints."""
+    if any(k.startswith("module.") for k in sd):
+        sd = {k[len("module."):] if k.startswith("module.") else k: v
+              for k, v in sd.items()}
+    model.load_state_dict(sd)
+
+
+def model_forward(model: nn.Module, model_name: str, data, device):
+    """Per-model forward dispatch (reference utils/train.py:63-90)."""
+    kw = dict(rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
+              counts_global=getattr(data, "counts_global", None))
+    if model_name in ("FastEGNN", "FastSchNet"):
+        node_attr = None if model.node_attr_nf == 0 else data.attr
+        return model(data.x, data.pos, data.vel, data.loc_mean,
+                     data.edge_index, data.batch, edge_attr=data.edge_attr,
+                     node_attr=node_attr, **kw)
+    if model_name == "FastTFN":
+        node_attr = None if model.node_attr_nf == 0 else data.attr
+        return model(data.x, data.pos, data.vel, data.loc_mean,
+                     data.edge_index, data.batch, data.attr,
+                     edge_attr=data.edge_attr, node_attr=node_attr, **kw)
+    if model_name == "FastRF":
+        return model(data.pos, data.vel, data.loc_mean, data.edge_index,
+                     data.batch, data.edge_attr, **kw)
+    if model_name == "SchNet":
+        return model(z=data.x, pos=data.pos, batch=data.batch,
+                     edge_index=data.edge_index), None
+    if model_name == "EGNN":
+        pred, _, _ = model(data.pos, data.x, data.edge_index, data.edge_attr,
+                           data.vel)
+        return pred, None
+    if model_name in ("RF", "RF_vel"):
+        pred = model(data.vel.norm(dim=-1, keepdim=True), data.pos,
+                     data.edge_index, data.vel, data.edge_attr)
+        return pred, None
+    if model_name in ("TFN", "OurDynamics"):
+        return model(data.pos, data.vel, data.attr, data.edge_index), None
+    if model_name in ("Linear", "Linear_dynamics"):
+        return model(data.pos, data.vel), None
+    raise NotImplementedError(f"{model_name} not implemented!")
+
+
+def _is_fast_model(model_name: str) -> bool:
+    return model_name.startswith("Fast")
+
+
+def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
+                       loss_fn, dataset_name, train_config, epoch_index, tag,
+                       subgraphs, world_size, device, grad_bucket=None,
+                       autocast_dtype=None, debug_lockstep=False,
+                       progress=True, step_timer=None):
+    backprop = tag == "train"
+    if backprop:
+        model.train()
+        optimizer.zero_grad(set_to_none=False)
+    else:
+        model.eval()
+
+    loss_accum = torch.zeros((), device=device)     # sum of weighted losses
+    counter = torch.zeros((), device=device)
+
+    iterator = enumerate(loader)
+    if progress and tqdm is not None:
+        iterator = tqdm(iterator, total=len(loader),
+                        desc=f"Epoch {epoch_index} - {tag.capitalize()} "
+                             f"[GPU {rank}]",
+                        position=rank, leave=False)
+
+    for step, data in iterator:
+        if step_timer is not None:
+            step_timer.start()
+        batch_size = data.num_graphs                # host-known, no sync
+        data = data.to(device)
+
+        if debug_lockstep and world_size > 1:
+            gathered = [torch.zeros_like(data.loc_mean)
+                        for _ in range(world_size)]
+            dist.all_gather(gathered, data.loc_mean)
+            if rank == 0:
+                for i in range(1, world_size):
+                    assert torch.allclose(gathered[0], gathered[i],
+                                          atol=1e-6), \
+                        "train loader out of lockstep across ranks"
+
+        # one collective per step: per-graph global node counts
+        data.counts_global = comm.global_counts(data.counts) \
+            if world_size > 1 else data.counts
+        total_node_cnt = data.counts_global.sum()
+        node_cnt = float(data.num_nodes)
+
+        with contextlib.ExitStack() as stack:
+            if not backprop:
+                stack.enter_context(torch.no_grad())
+            if autocast_dtype is not None:
+                stack.enter_context(torch.autocast("cuda",
+                                                   dtype=autocast_dtype))
+            loc_pred, virtual_node_loc = model_forward(
+                model, model_name, data, device)
+
+        loss_loc = loss_fn(loc_pred.float(), data.target)
+        # node-count weighting: per-rank share of the global per-node MSE
+        weight = node_cnt / total_node_cnt
+        loss_loc = weight * loss_loc
+        loss_accum = loss_accum + loss_loc.detach() * batch_size
+        counter = counter + batch_size
+
+        if _is_fast_model(model_name) and virtual_node_loc is not None:
+            vloc = virtual_node_loc.permute(0, 2, 1).float()  # [B, C, 3]
+            lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
+                          data.counts, train_config.mmd.sigma,
+                          train_config.mmd.samples)
+            loss_loc = loss_loc + train_config.mmd.weight * weight * lm
+
+        if backprop:
+            (loss_loc / float(train_config.accumulation_steps)).backward()
+            if (step + 1) % train_config.accumulation_steps == 0:
+                if grad_bucket is not None:
+                    grad_bucket.sync()              # SUM == ref avg * ws
+                if ((world_size > 1 or dataset_name == "LargeFluid")
+                        and model_name == "FastEGNN"):
+                    nn.utils.clip_grad_norm_(model.parameters(), max_norm=0.3)
+                optimizer.step()
+                if scheduler is not None:
+                    scheduler.step()
+                optimizer.zero_grad(set_to_none=False)
+        if step_timer is not None:
+            step_timer.stop()
+
+    # deferred logging reduce: one collective per epoch (ref: per step)
+    if world_size > 1:
+        dist.all_reduce(loss_accum, op=dist.ReduceOp.SUM)
+    avg = (loss_accum / counter).item()
+    if rank == 0:
+        prefix = "" if backprop else "==> "
+        print(f"{prefix}{tag} epoch: {epoch_index}, avg loss: {avg:.5f}")
+    return avg
+
+
+def train(rank, model, model_name, optimizer, scheduler, loader_train,
+          loader_valid, loader_test, train_config, log_config, config,
+          start_epoch, device=None, grad_bucket: Optional[GradBucket] = None,
+          autocast_dtype=None, progress=True):
+    world_size = config.data.world_size
+    device = device if device is not None else (
+        torch.device(f"cuda:{rank}") if torch.cuda.is_available()
+        else torch.device("cpu"))
+    loss_mse = nn.MSELoss()
+    debug_lockstep = bool(train_config.get("debug_lockstep", False) or
+                          os.environ.get("DISTEGNN_DEBUG_LOCKSTEP") == "1")
+
+    log_dict = {"epochs": [], "loss": [], "loss_train": []}
+    best_log_dict = {"epoch_index": 0, "loss_valid": 1e8, "loss_test": 1e8,
+                     "loss_train": 1e8}
+    if rank == 0:
+        log_dir = os.path.join(log_config.log_dir, log_config.exp_name, "log")
+        os.makedirs(log_dir, exist_ok=True)
+        state_dict_dir = os.path.join(log_config.log_dir, log_config.exp_name,
+                                      "state_dict")
+        os.makedirs(state_dict_dir, exist_ok=True)
+        start = time.perf_counter()
+
+    early_stop_flag = torch.tensor(0, device=device)
+
+    for epoch_index in range(1 + start_epoch, train_config.epochs + 1):
+        if early_stop_flag.item() == 1:
+            print(f"Device {rank} stop succeed!")
+            break
+
+        loss_train = train_single_epoch(
+            rank, model, model_name, loader_train, optimizer, scheduler,
+            loss_mse, config.data.dataset_name, train_config, epoch_index,
+            tag="train", subgraphs=config.model.virtual_channels,
+            world_size=world_size, device=device, grad_bucket=grad_bucket,
+            autocast_dtype=autocast_dtype, debug_lockstep=debug_lockstep,
+            progress=progress)
+        if rank == 0:
+            log_dict["loss_train"].append(loss_train)
+
+        if epoch_index % log_config.test_interval == 0:
+            loss_valid = train_single_epoch(
+                rank, model, model_name, loader_valid, optimizer, scheduler,
+                loss_mse, config.data.dataset_name, train_config, epoch_index,
+                tag="valid", subgraphs=config.model.virtual_channels,
+                world_size=world_size, device=device,
+                autocast_dtype=autocast_dtype, progress=progress)
+            loss_test = train_single_epoch(
+                rank, model, model_name, loader_test, optimizer, scheduler,
+                loss_mse, config.data.dataset_name, train_config, epoch_index,
+                tag="test", subgraphs=config.model.virtual_channels,
+                world_size=world_size, device=device,
+                autocast_dtype=autocast_dtype, progress=progress)
+
+            if rank == 0:
+                log_dict["epochs"].append(epoch_index)
+                log_dict["loss"].append(loss_test)
+                state = {
+                    "epoch": epoch_index,
+                    "model_state_dict": state_dict_for_save(model, world_size),
+                    "optimizer_state_dict": optimizer.state_dict(),
+                    "scheduler_state_dict": (None if scheduler is None
+                                             else scheduler.state_dict()),
+                    "loss_train": loss_train, "loss_valid": loss_valid,
+                    "loss_test": loss_test,
+                    "config": config.to_dict() if hasattr(config, "to_dict")
+                    else config,
+                }
+                if loss_valid < best_log_dict["loss_valid"]:
+                    best_log_dict = {"epoch_index": epoch_index,
+                                     "loss_valid": loss_valid,
+                                     "loss_test": loss_test,
+                                     "loss_train": loss_train}
+                    torch.save(state, os.path.join(state_dict_dir,
+                                                   "best_model.pth"))
+                print(f"*** Best Valid Loss: {best_log_dict['loss_valid']:.5f}"
+                      f" | Best Test Loss: {best_log_dict['loss_test']:.5f}"
+                      f" | Best Epoch Index: {best_log_dict['epoch_index']}")
+                torch.save(state, os.path.join(state_dict_dir,
+                                               "last_model.pth"))
+
+            if rank == 0 and (epoch_index - best_log_dict["epoch_index"]
+                              >= train_config.early_stop):
+                best_log_dict["early_stop"] = epoch_index
+                print(f"Early stopped! Epoch: {epoch_index}")
+                early_stop_flag.fill_(1)
+            if world_size > 1:
+                dist.all_reduce(early_stop_flag, op=dist.ReduceOp.MAX)
+
+        if rank == 0:
+            best_log_dict["time_cost"] = time.perf_counter() - start
+            payload = [best_log_dict, log_dict,
+                       config.to_dict() if hasattr(config, "to_dict")
+                       else config]
+            with open(os.path.join(log_dir, "log.json"), "w") as f:
+                f.write(json.dumps(payload, indent=4))
+
+    if world_size > 1:
+        dist.barrier()
+        comm.destroy()
+    if rank == 0:
+        return best_log_dict, log_dict

@@ -1,0 +1,85 @@
+import torch
+
+from distegnn_amd.data.graph import Batch, Data, collate, sort_edges_by_row
+from distegnn_amd.data.partition import (SPLITTERS, graph_partition,
+                                         split_large_graph_random)
+from distegnn_amd.data.synthetic import (WORKLOADS, make_cloud_sample,
+                                         make_cutoff_dataset,
+                                         make_distributed_dataset)
+from distegnn_amd.ops import reference as R
+
+
+def test_sort_edges():
+    ei = torch.tensor([[3, 1, 2, 1], [0, 2, 1, 0]])
+    ea = torch.arange(4).float().unsqueeze(-1)
+    ei2, ea2 = sort_edges_by_row(ei, ea)
+    assert ei2[0].tolist() == [1, 1, 2, 3]
+    # stable: original order kept within row groups
+    assert ea2.squeeze(-1).tolist() == [1.0, 3.0, 2.0, 0.0]
+
+
+def test_collate_offsets_and_csr(small_batch):
+    b = small_batch
+    assert b.num_graphs == 3
+    assert b.ptr.tolist() == [0, 100, 200, 300]
+    assert b.batch.shape == (300,)
+    assert b.counts.tolist() == [100.0, 100.0, 100.0]
+    # rowptr consistent with edge rows
+    row = b.edge_index[0]
+    deg = torch.bincount(row, minlength=300)
+    assert torch.equal(b.rowptr[1:] - b.rowptr[:-1], deg)
+    assert torch.all(row[1:] >= row[:-1])
+    # edges stay within their graph
+    src_g = b.batch[b.edge_index[0]]
+    dst_g = b.batch[b.edge_index[1]]
+    assert torch.equal(src_g, dst_g)
+
+
+def test_random_split_balance_and_loc_mean():
+    g = torch.Generator().manual_seed(0)
+    n = 503
+    pos = torch.rand(n, 3, generator=g)
+    x = torch.rand(n, 2, generator=g)
+    parts = split_large_graph_random(pos, x, pos.clone(), pos.clone(),
+                                     x[:, :1], 0.2, 4, generator=g)
+    sizes = [p.num_nodes for p in parts]
+    assert sum(sizes) == n
+    assert max(sizes) - min(sizes) <= 3
+    for p in parts:
+        assert torch.allclose(p.loc_mean, pos.mean(0, keepdim=True))
+        assert p.edge_attr.shape == (p.num_edges, 2)
+
+
+def test_graph_partition_balanced_and_local():
+    g = torch.Generator().manual_seed(1)
+    pos = torch.rand(400, 3, generator=g)
+    ei = R.radius_graph(pos, 0.25)
+    labels = graph_partition(ei, 400, 4, pos=pos)
+    sizes = torch.bincount(labels, minlength=4)
+    assert sizes.min() >= 90 and sizes.max() <= 110
+    # locality: cut fraction well below the random-partition expectation (75%)
+    cut = (labels[ei[0]] != labels[ei[1]]).float().mean()
+    assert cut < 0.5
+
+
+def test_kmeans_split_mode():
+    data = make_distributed_dataset("Water-3D", 1, 2, split_mode="kmeans",
+                                    seed=0, n_override=400)
+    assert len(data) == 2 and len(data[0]) == 1
+    assert data[0][0].num_nodes + data[1][0].num_nodes == 400
+
+
+def test_synthetic_density():
+    """Synthetic clouds reproduce the published average degree ballpark."""
+    rng = torch.Generator().manual_seed(0)
+    s = make_cloud_sample("Water-3D", rng)
+    ei = R.radius_graph(s["pos"], s["radius"])
+    avg_deg = ei.size(1) / s["pos"].size(0)
+    assert 8 <= avg_deg <= 18  # published ~12.2
+
+
+def test_synthetic_fields_fluid():
+    rng = torch.Generator().manual_seed(0)
+    s = make_cloud_sample("Fluid113K", rng, n_override=1000)
+    assert s["x"].shape == (1000, 3)      # [visc, mass, |v|]
+    assert s["attr"].shape == (1000, 2)   # [visc, mass]
