@@ -1,0 +1,202 @@
+"""Flagship benchmark: LargeFluid-113K DistEGNN training step on MI355X.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W`` runs the
+headline training workload (BASELINE.json: train step time (ms) for
+LargeFluid-113K DistEGNN) on N GPUs of one node; for N>1 it is launched by
+``python -m torch.distributed.run --nproc-per-node N ... bench.py`` with one
+rank per GPU over RCCL. Rank 0 prints ONE JSON line.
+
+Workload: synthetic 113,140-node fluid cloud at the published density
+(~1.7M edges at r=0.075), random-init FastEGNN (H=64, L=4, C=5),
+partitioned across ranks (split_mode random by default, --split-mode metis
+supported), grad-accumulation 4 as in config/largefluid_distegnn.yaml.
+STRONG scaling: the 113K-node graph is fixed; more GPUs = smaller
+partitions per rank + the virtual-node all-reduce.
+
+A timed step = H2D of the batch + forward + coord-MSE + MMD + backward
+(+ gradient all-reduce, clip and Adam step every 4th step). Compute dtype
+bf16 (autocast) for MLP GEMMs; coordinates/reductions fp32.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from distegnn_amd.data.graph import collate
+from distegnn_amd.data.synthetic import make_cloud_sample
+from distegnn_amd.data.partition import SPLITTERS
+from distegnn_amd.models import FastEGNN
+from distegnn_amd.parallel import comm
+from distegnn_amd.parallel.comm import GradBucket
+from distegnn_amd.runtime.losses import mmd_loss
+from distegnn_amd.utils import fix_seed
+
+
+def build_rank_batches(rank, world_size, num_batches, n_nodes, radius,
+                       split_mode, seed):
+    """Identical clouds on every rank (fixed seed); each rank keeps its
+    partition — mirrors the reference's offline preprocessing."""
+    batches = []
+    rng = torch.Generator().manual_seed(seed)
+    for _ in range(num_batches):
+        s = make_cloud_sample("Fluid113K", rng, n_override=n_nodes)
+        if world_size == 1:
+            parts = SPLITTERS["random"](
+                pos=s["pos"], x=s["x"], target=s["target"], vel=s["vel"],
+                attr=s["attr"], radius=radius, world_size=1, device="cpu",
+                generator=rng)
+        elif split_mode == "random":
+            parts = SPLITTERS["random"](
+                pos=s["pos"], x=s["x"], target=s["target"], vel=s["vel"],
+                attr=s["attr"], radius=radius, world_size=world_size,
+                device="cpu", generator=rng)
+        else:
+            parts = SPLITTERS[split_mode](
+                pos=s["pos"], x=s["x"], target=s["target"], vel=s["vel"],
+                attr=s["attr"], outer_radius=radius, inner_radius=radius,
+                world_size=world_size, device="cpu")
+        batches.append(collate([parts[rank]]))
+    return batches
+
+
+def train_step(model, batch, optimizer, grad_bucket, step, accum,
+               mmd_sigma, mmd_samples, world_size, device, autocast_dtype):
+    data = batch.to(device)
+    data.counts_global = (comm.global_counts(data.counts)
+                          if world_size > 1 else data.counts)
+    total_node_cnt = data.counts_global.sum()
+    chunks = None
+    if getattr(data, "pool_chunk_begin", None) is not None:
+        chunks = (data.pool_chunk_begin, data.pool_chunk_end,
+                  data.pool_seg_chunk_ptr)
+    ctx = (torch.autocast("cuda", dtype=autocast_dtype)
+           if autocast_dtype is not None else torch.enable_grad())
+    with ctx:
+        loc_pred, vloc = model(
+            data.x, data.pos, data.vel, data.loc_mean, data.edge_index,
+            data.batch, edge_attr=data.edge_attr, node_attr=data.attr,
+            rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
+            counts_global=data.counts_global, pool_chunks=chunks)
+    loss = torch.nn.functional.mse_loss(loc_pred.float(), data.target)
+    weight = float(data.num_nodes) / total_node_cnt
+    loss = weight * loss
+    mse_log = loss.detach()
+    lm = mmd_loss(vloc.permute(0, 2, 1).float(), data.target, data.batch,
+                  data.ptr, data.counts, mmd_sigma, mmd_samples)
+    loss = loss + 0.01 * weight * lm
+    (loss / accum).backward()
+    if (step + 1) % accum == 0:
+        if grad_bucket is not None:
+            grad_bucket.sync()
+        torch.nn.utils.clip_grad_norm_(model.parameters(), max_norm=0.3)
+        optimizer.step()
+        optimizer.zero_grad(set_to_none=False)
+    return mse_log
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=16)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--nodes", type=int, default=113140)
+    ap.add_argument("--radius", type=float, default=0.075)
+    ap.add_argument("--split-mode", type=str, default="random",
+                    choices=["random", "metis", "kmeans"])
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["bf16", "fp32"])
+    ap.add_argument("--virtual-channels", type=int, default=5)
+    ap.add_argument("--num-batches", type=int, default=2,
+                    help="distinct synthetic samples to cycle through")
+    args = ap.parse_args()
+
+    rank, world_size = comm.init_distributed()
+    if world_size != args.gpus and int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        args.gpus = world_size
+    world_size = max(world_size, 1)
+    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    device = torch.device(f"cuda:{rank}")
+    torch.cuda.set_device(device)
+
+    fix_seed(43)
+    t_data = time.perf_counter()
+    batches = build_rank_batches(rank, world_size, args.num_batches,
+                                 args.nodes, args.radius, args.split_mode,
+                                 seed=43)
+    if rank == 0:
+        print(f"# data built in {time.perf_counter() - t_data:.1f}s: "
+              f"{batches[0].num_nodes} nodes/rank, "
+              f"{batches[0].num_edges} edges/rank", flush=True)
+
+    model = FastEGNN(node_feat_nf=3, node_attr_nf=2, edge_attr_nf=2,
+                     hidden_nf=64, virtual_channels=args.virtual_channels,
+                     world_size=world_size, n_layers=4).to(device)
+    grad_bucket = None
+    if world_size > 1:
+        grad_bucket = GradBucket(model)
+        grad_bucket.broadcast_parameters()
+    optimizer = torch.optim.Adam(model.parameters(), lr=5e-4,
+                                 weight_decay=1e-12)
+    autocast_dtype = torch.bfloat16 if args.dtype == "bf16" else None
+    accum = 4
+    mmd_sigma, mmd_samples = 3.0, 50
+    model.train()
+
+    mse = None
+    for w in range(args.warmup):
+        mse = train_step(model, batches[w % len(batches)], optimizer,
+                         grad_bucket, w, accum, mmd_sigma, mmd_samples,
+                         world_size, device, autocast_dtype)
+    comm.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for k in range(args.steps):
+        mse = train_step(model, batches[k % len(batches)], optimizer,
+                         grad_bucket, args.warmup + k, accum, mmd_sigma,
+                         mmd_samples, world_size, device, autocast_dtype)
+    comm.barrier()
+    torch.cuda.synchronize()
+    elapsed = torch.tensor(time.perf_counter() - t0, device=device)
+    if world_size > 1:
+        torch.distributed.all_reduce(elapsed,
+                                     op=torch.distributed.ReduceOp.MAX)
+    ms_per_step = elapsed.item() * 1000.0 / args.steps
+
+    if rank == 0:
+        out = {
+            "metric": "train_step_time_ms",
+            "value": ms_per_step,
+            "unit": "ms",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "FastEGNN-DistEGNN",
+                "dataset": "LargeFluid-113K",
+                "global_batch": 1,
+                "nodes": args.nodes,
+                "radius": args.radius,
+                "split_mode": args.split_mode,
+                "hidden_nf": 64,
+                "n_layers": 4,
+                "virtual_channels": args.virtual_channels,
+                "accumulation_steps": accum,
+                "parallelism": f"graph-partition dp{world_size}",
+                "coord_mse": float(mse.item()) if mse is not None else None,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    comm.destroy()
+
+
+if __name__ == "__main__":
+    main()

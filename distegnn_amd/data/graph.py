@@ -145,6 +145,28 @@ class Batch:
         )
         self.counts = (self.ptr[1:] - self.ptr[:-1]).to(torch.float32)
 
+        # Precomputed chunk tables for the HIP two-stage graph pooling when
+        # per-graph node blocks are huge (DistEGNN: one 100K+ node graph per
+        # rank). Built HERE on the host (ptr is host-known) so the training
+        # hot loop never syncs. See ops/csrc/segment_reduce.hip.
+        max_seg = int((self.ptr[1:] - self.ptr[:-1]).max()) if b > 0 else 0
+        if max_seg > 4096:
+            chunk = 2048
+            cb, ce, scp = [], [], [0]
+            for i in range(b):
+                s, e = int(self.ptr[i]), int(self.ptr[i + 1])
+                for k in range(s, e, chunk):
+                    cb.append(k)
+                    ce.append(min(k + chunk, e))
+                scp.append(len(cb))
+            self.pool_chunk_begin = torch.tensor(cb, dtype=torch.long)
+            self.pool_chunk_end = torch.tensor(ce, dtype=torch.long)
+            self.pool_seg_chunk_ptr = torch.tensor(scp, dtype=torch.long)
+        else:
+            self.pool_chunk_begin = None
+            self.pool_chunk_end = None
+            self.pool_seg_chunk_ptr = None
+
     @property
     def num_nodes(self) -> int:
         return int(self.ptr[-1])

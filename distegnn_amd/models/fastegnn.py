@@ -127,7 +127,7 @@ class EGCLVel(nn.Module):
     def forward(self, h, edge_index, coord, vel, virtual_coord, virtual_feat,
                 batch, edge_attr=None, node_attr=None, *,
                 rowptr=None, ptr=None, counts=None, counts_global=None,
-                num_graphs=None):
+                num_graphs=None, pool_chunks=None):
         """virtual_coord: [B, C, 3]; virtual_feat: [B, C, H] (channels-major).
 
         rowptr/ptr/counts/counts_global come from the Batch (device-side);
@@ -154,7 +154,8 @@ class EGCLVel(nn.Module):
             edge_feat = edge_feat * self.att_mlp(edge_feat)
 
         # --- global coord mean (site A collective) ----------------------
-        coord_mean = ops.graph_mean_pool(coord, batch, b, ptr=ptr, counts=counts)
+        coord_mean = ops.graph_mean_pool(coord, batch, b, ptr=ptr, counts=counts,
+                                         chunks=pool_chunks)
         if dist_active:
             coord_mean = comm.fused_weighted_average_reduce(
                 [coord_mean], counts, counts_global)
@@ -192,10 +193,12 @@ class EGCLVel(nn.Module):
         # --- virtual aggregates (fused site B+C collective) -------------
         trans_x = vdiff * self.coord_mlp_v_virtual(v_msg)      # [N, C, 3]
         agg_vc = ops.graph_mean_pool(
-            trans_x.reshape(n, -1), batch, b, ptr=ptr, counts=counts
+            trans_x.reshape(n, -1), batch, b, ptr=ptr, counts=counts,
+            chunks=pool_chunks
         ).reshape(b, c, 3)
         agg_vf = ops.graph_mean_pool(
-            v_msg.reshape(n, -1), batch, b, ptr=ptr, counts=counts
+            v_msg.reshape(n, -1), batch, b, ptr=ptr, counts=counts,
+            chunks=pool_chunks
         ).reshape(b, c, self.hidden_nf)
         if dist_active:
             agg_vc, agg_vf = comm.fused_weighted_average_reduce(
@@ -254,7 +257,8 @@ class FastEGNN(nn.Module):
 
     def forward(self, node_feat, node_loc, node_vel, loc_mean, edge_index,
                 data_batch, edge_attr=None, node_attr=None, *,
-                rowptr=None, ptr=None, counts=None, counts_global=None):
+                rowptr=None, ptr=None, counts=None, counts_global=None,
+                pool_chunks=None):
         """Returns (loc_pred [N, 3], virtual_node_loc [B, 3, C]).
 
         API parity with the reference forward (FastEGNN.py:296-307); the
@@ -280,5 +284,6 @@ class FastEGNN(nn.Module):
                 h, edge_index, loc, node_vel, virtual_loc, virtual_feat,
                 data_batch, edge_attr=edge_attr, node_attr=node_attr,
                 rowptr=rowptr, ptr=ptr, counts=counts,
-                counts_global=counts_global, num_graphs=b)
+                counts_global=counts_global, num_graphs=b,
+                pool_chunks=pool_chunks)
         return loc, virtual_loc.transpose(1, 2)  # [B, 3, C] API parity

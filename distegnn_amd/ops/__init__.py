@@ -102,14 +102,21 @@ def segment_mean(data: torch.Tensor, row: torch.Tensor, num_segments: int,
 
 
 class _GraphPoolFn(torch.autograd.Function):
-    """Per-graph sum/mean pooling over ptr-delimited node blocks."""
+    """Per-graph sum/mean pooling over ptr-delimited node blocks.
+
+    ``chunks`` (chunk_begin, chunk_end, seg_chunk_ptr) — host-precomputed at
+    collate time (Batch) — routes huge segments through the two-stage
+    deterministic kernel (one 100K+-node graph per rank in DistEGNN)."""
 
     @staticmethod
-    def forward(ctx, x, batch, ptr, num_graphs, mean):
+    def forward(ctx, x, batch, ptr, num_graphs, mean, chunks=None):
         ext = _require_ext("graph_pool")
         if ext is None:
             out = (reference.graph_mean_pool if mean else reference.graph_sum_pool)(
                 x, batch, num_graphs)
+        elif chunks is not None:
+            cb, ce, scp = chunks
+            out = ext.segment_reduce_chunked(x, ptr, cb, ce, scp, bool(mean))
         else:
             out = ext.segment_reduce_csr(x, ptr, bool(mean))
         ctx.save_for_backward(batch, ptr)
@@ -124,21 +131,25 @@ class _GraphPoolFn(torch.autograd.Function):
             cnt = (ptr[1:] - ptr[:-1]).clamp(min=1).to(gout.dtype)
             gout = gout / cnt.view(-1, *([1] * (gout.dim() - 1)))
         gx = gout.index_select(0, batch)
-        return gx, None, None, None, None
+        return gx, None, None, None, None, None
 
 
 def graph_sum_pool(x: torch.Tensor, batch: torch.Tensor, num_graphs: int,
-                   ptr: Optional[torch.Tensor] = None) -> torch.Tensor:
+                   ptr: Optional[torch.Tensor] = None,
+                   chunks=None) -> torch.Tensor:
     if x.is_cuda and ptr is not None:
-        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs, False)
+        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs,
+                                  False, chunks)
     return reference.graph_sum_pool(x, batch, num_graphs)
 
 
 def graph_mean_pool(x: torch.Tensor, batch: torch.Tensor, num_graphs: int,
                     ptr: Optional[torch.Tensor] = None,
-                    counts: Optional[torch.Tensor] = None) -> torch.Tensor:
+                    counts: Optional[torch.Tensor] = None,
+                    chunks=None) -> torch.Tensor:
     if x.is_cuda and ptr is not None:
-        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs, True)
+        return _GraphPoolFn.apply(x.contiguous(), batch, ptr, num_graphs,
+                                  True, chunks)
     return reference.graph_mean_pool(x, batch, num_graphs, counts=counts)
 
 

@@ -61,8 +61,13 @@ def load_state_dict_compat(model: nn.Module, sd: dict):
 
 def model_forward(model: nn.Module, model_name: str, data, device):
     """Per-model forward dispatch (reference utils/train.py:63-90)."""
+    chunks = None
+    if getattr(data, "pool_chunk_begin", None) is not None:
+        chunks = (data.pool_chunk_begin, data.pool_chunk_end,
+                  data.pool_seg_chunk_ptr)
     kw = dict(rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
-              counts_global=getattr(data, "counts_global", None))
+              counts_global=getattr(data, "counts_global", None),
+              pool_chunks=chunks)
     if model_name in ("FastEGNN", "FastSchNet"):
         node_attr = None if model.node_attr_nf == 0 else data.attr
         return model(data.x, data.pos, data.vel, data.loc_mean,
