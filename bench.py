@@ -79,7 +79,8 @@ def train_step(model, batch, optimizer, grad_bucket, step, accum,
             data.x, data.pos, data.vel, data.loc_mean, data.edge_index,
             data.batch, edge_attr=data.edge_attr, node_attr=data.attr,
             rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
-            counts_global=data.counts_global, pool_chunks=chunks)
+            counts_global=data.counts_global, pool_chunks=chunks,
+            colptr=data.colptr, col_perm=data.col_perm)
     loss = torch.nn.functional.mse_loss(loc_pred.float(), data.target)
     weight = float(data.num_nodes) / total_node_cnt
     loss = weight * loss

@@ -145,6 +145,16 @@ class Batch:
         )
         self.counts = (self.ptr[1:] - self.ptr[:-1]).to(torch.float32)
 
+        # Column-side CSR for gather backward: sorting permutation of the
+        # edge destinations (edge_index[1]) + its pointer. Lets the HIP path
+        # run scatter-free segment sums in the backward of h[col] gathers
+        # (ops._GatherRowsFn).
+        col = self.edge_index[1]
+        self.col_perm = torch.argsort(col, stable=True)
+        self.colptr = torch.cat(
+            [col.new_zeros(1), torch.cumsum(torch.bincount(col, minlength=n), 0)]
+        )
+
         # Precomputed chunk tables for the HIP two-stage graph pooling when
         # per-graph node blocks are huge (DistEGNN: one 100K+ node graph per
         # rank). Built HERE on the host (ptr is host-known) so the training

@@ -114,9 +114,11 @@ class EGCLVel(nn.Module):
         )
 
     # --- geometry -------------------------------------------------------
-    def coord2radial(self, edge_index, coord):
+    def coord2radial(self, edge_index, coord, rowptr=None, colptr=None,
+                     col_perm=None):
         row, col = edge_index[0], edge_index[1]
-        coord_diff = coord.index_select(0, row) - coord.index_select(0, col)
+        coord_diff = (ops.gather_rows(coord, row, rowptr)
+                      - ops.gather_rows(coord, col, colptr, col_perm))
         radial = coord_diff.pow(2).sum(dim=1, keepdim=True)
         if self.normalize:
             norm = radial.sqrt().detach() + self.epsilon
@@ -127,7 +129,8 @@ class EGCLVel(nn.Module):
     def forward(self, h, edge_index, coord, vel, virtual_coord, virtual_feat,
                 batch, edge_attr=None, node_attr=None, *,
                 rowptr=None, ptr=None, counts=None, counts_global=None,
-                num_graphs=None, pool_chunks=None):
+                num_graphs=None, pool_chunks=None, colptr=None,
+                col_perm=None):
         """virtual_coord: [B, C, 3]; virtual_feat: [B, C, H] (channels-major).
 
         rowptr/ptr/counts/counts_global come from the Batch (device-side);
@@ -138,17 +141,21 @@ class EGCLVel(nn.Module):
         row = edge_index[0]
         dist_active = self.world_size > 1 and comm.is_distributed()
 
-        radial, coord_diff = self.coord2radial(edge_index, coord)
+        radial, coord_diff = self.coord2radial(edge_index, coord,
+                                               rowptr=rowptr, colptr=colptr,
+                                               col_perm=col_perm)
         # (X_c - x_i): [N, C, 3]; its norm: [N, C, 1]
-        vdiff = virtual_coord.index_select(0, batch) - coord.unsqueeze(1)
+        vdiff = (ops.gather_rows(virtual_coord, batch, ptr,
+                                 chunks=pool_chunks)
+                 - coord.unsqueeze(1))
         vradial = vdiff.norm(p=2, dim=-1, keepdim=True)
 
         # --- edge model (real-real), phi_e -----------------------------
+        h_row = ops.gather_rows(h, row, rowptr)
+        h_col = ops.gather_rows(h, edge_index[1], colptr, col_perm)
         edge_in = torch.cat(
-            [h.index_select(0, row), h.index_select(0, edge_index[1]),
-             radial, edge_attr] if edge_attr is not None else
-            [h.index_select(0, row), h.index_select(0, edge_index[1]), radial],
-            dim=1)
+            [h_row, h_col, radial, edge_attr] if edge_attr is not None
+            else [h_row, h_col, radial], dim=1)
         edge_feat = self.edge_mlp(edge_in)
         if self.attention:
             edge_feat = edge_feat * self.att_mlp(edge_feat)
@@ -167,9 +174,9 @@ class EGCLVel(nn.Module):
         # --- virtual edge model, phi_ev: [N, C, 2H+1+C] -> [N, C, H] ----
         v_in = torch.cat([
             h.unsqueeze(1).expand(n, c, h.size(1)),
-            virtual_feat.index_select(0, batch),
+            ops.gather_rows(virtual_feat, batch, ptr, chunks=pool_chunks),
             vradial,
-            gram.index_select(0, batch),
+            ops.gather_rows(gram, batch, ptr, chunks=pool_chunks),
         ], dim=-1)
         v_msg = self.edge_mlp_virtual(v_in)
         if self.attention:
@@ -258,7 +265,7 @@ class FastEGNN(nn.Module):
     def forward(self, node_feat, node_loc, node_vel, loc_mean, edge_index,
                 data_batch, edge_attr=None, node_attr=None, *,
                 rowptr=None, ptr=None, counts=None, counts_global=None,
-                pool_chunks=None):
+                pool_chunks=None, colptr=None, col_perm=None):
         """Returns (loc_pred [N, 3], virtual_node_loc [B, 3, C]).
 
         API parity with the reference forward (FastEGNN.py:296-307); the
@@ -285,5 +292,5 @@ class FastEGNN(nn.Module):
                 data_batch, edge_attr=edge_attr, node_attr=node_attr,
                 rowptr=rowptr, ptr=ptr, counts=counts,
                 counts_global=counts_global, num_graphs=b,
-                pool_chunks=pool_chunks)
+                pool_chunks=pool_chunks, colptr=colptr, col_perm=col_perm)
         return loc, virtual_loc.transpose(1, 2)  # [B, 3, C] API parity

@@ -67,7 +67,8 @@ def model_forward(model: nn.Module, model_name: str, data, device):
                   data.pool_seg_chunk_ptr)
     kw = dict(rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
               counts_global=getattr(data, "counts_global", None),
-              pool_chunks=chunks)
+              pool_chunks=chunks, colptr=getattr(data, "colptr", None),
+              col_perm=getattr(data, "col_perm", None))
     if model_name in ("FastEGNN", "FastSchNet"):
         node_attr = None if model.node_attr_nf == 0 else data.attr
         return model(data.x, data.pos, data.vel, data.loc_mean,

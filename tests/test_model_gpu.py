@@ -30,7 +30,7 @@ def make(n_layers=2, hidden=32, c=3, feat=2, attr=0):
 def fwd(m, b, **kw):
     return m(b.x, b.pos, b.vel, b.loc_mean, b.edge_index, b.batch,
              edge_attr=b.edge_attr, rowptr=b.rowptr, ptr=b.ptr,
-             counts=b.counts, **kw)
+             counts=b.counts, colptr=b.colptr, col_perm=b.col_perm, **kw)
 
 
 def test_gpu_matches_cpu_fp32():

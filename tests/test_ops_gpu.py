@@ -128,6 +128,37 @@ def test_radius_graph_dispatch_path():
     assert ei.is_cuda and ei.size(0) == 2
 
 
+def test_gather_rows_backward_row_sorted():
+    """gather_rows backward (CSR segment sum) == index_select backward."""
+    m, n, f = 4000, 300, 48
+    row, rowptr = make_csr(m, n, seed=4)
+    x = torch.randn(n, f, requires_grad=True)
+    xg = x.detach().clone().to(dev()).requires_grad_(True)
+    y = x.index_select(0, row)
+    (y * torch.linspace(0.5, 1.5, m).unsqueeze(-1)).sum().backward()
+    yg = ops.gather_rows(xg, row.to(dev()), rowptr.to(dev()))
+    (yg * torch.linspace(0.5, 1.5, m, device=dev()).unsqueeze(-1)).sum() \
+        .backward()
+    assert torch.allclose(yg.cpu(), y, atol=1e-5)
+    assert torch.allclose(xg.grad.cpu(), x.grad, atol=1e-4, rtol=1e-4)
+
+
+def test_gather_rows_backward_unsorted_with_perm():
+    m, n, f = 4000, 300, 16
+    g = torch.Generator().manual_seed(5)
+    col = torch.randint(0, n, (m,), generator=g)
+    perm = torch.argsort(col, stable=True)
+    colptr = torch.cat([torch.zeros(1, dtype=torch.long),
+                        torch.cumsum(torch.bincount(col, minlength=n), 0)])
+    x = torch.randn(n, f, requires_grad=True)
+    xg = x.detach().clone().to(dev()).requires_grad_(True)
+    w = torch.randn(m, 1, generator=g)
+    (x.index_select(0, col) * w).sum().backward()
+    (ops.gather_rows(xg, col.to(dev()), colptr.to(dev()),
+                     perm=perm.to(dev())) * w.to(dev())).sum().backward()
+    assert torch.allclose(xg.grad.cpu(), x.grad, atol=1e-4, rtol=1e-4)
+
+
 def test_segment_autograd_gpu():
     """ops.segment_mean custom Function: forward HIP, backward gather."""
     m, n, f = 3000, 200, 32
