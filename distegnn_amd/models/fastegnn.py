@@ -62,6 +62,7 @@ class EGCLVel(nn.Module):
         self.normalize = normalize
         self.coords_agg = coords_agg
         self.hidden_nf = hidden_nf
+        self.tanh = tanh
         self.world_size = world_size
         self.virtual_channels = virtual_channels
         self.epsilon = 1e-8
@@ -141,24 +142,41 @@ class EGCLVel(nn.Module):
         row = edge_index[0]
         dist_active = self.world_size > 1 and comm.is_distributed()
 
-        radial, coord_diff = self.coord2radial(edge_index, coord,
-                                               rowptr=rowptr, colptr=colptr,
-                                               col_perm=col_perm)
         # (X_c - x_i): [N, C, 3]; its norm: [N, C, 1]
         vdiff = (ops.gather_rows(virtual_coord, batch, ptr,
                                  chunks=pool_chunks)
                  - coord.unsqueeze(1))
         vradial = vdiff.norm(p=2, dim=-1, keepdim=True)
 
-        # --- edge model (real-real), phi_e -----------------------------
-        h_row = ops.gather_rows(h, row, rowptr)
-        h_col = ops.gather_rows(h, edge_index[1], colptr, col_perm)
-        edge_in = torch.cat(
-            [h_row, h_col, radial, edge_attr] if edge_attr is not None
-            else [h_row, h_col, radial], dim=1)
-        edge_feat = self.edge_mlp(edge_in)
-        if self.attention:
-            edge_feat = edge_feat * self.att_mlp(edge_feat)
+        # --- edge block (real-real): phi_e messages + phi_x translations -
+        # Fusable form (the standard FastEGNN configuration): one MFMA HIP
+        # kernel computes gather + phi_e MLP + phi_x head + d_ij * phi_x and
+        # returns the per-node segment MEANS directly (ops.fused_edge_block;
+        # csrc/fused_edge.hip). Non-standard variants (attention, tanh,
+        # sum aggregation, no edge_attr) take the composed eager path.
+        fuse = (not self.attention and not self.tanh
+                and self.coords_agg == "mean" and edge_attr is not None)
+        if fuse:
+            agg_e, agg = ops.fused_edge_block(
+                h, coord, edge_attr, row, edge_index[1], rowptr, colptr,
+                col_perm,
+                self.edge_mlp[0].weight, self.edge_mlp[0].bias,
+                self.edge_mlp[2].weight, self.edge_mlp[2].bias,
+                self.coord_mlp_r[0].weight, self.coord_mlp_r[0].bias,
+                self.coord_mlp_r[2].weight.reshape(-1),
+                self.normalize, self.epsilon)
+        else:
+            radial, coord_diff = self.coord2radial(
+                edge_index, coord, rowptr=rowptr, colptr=colptr,
+                col_perm=col_perm)
+            h_row = ops.gather_rows(h, row, rowptr)
+            h_col = ops.gather_rows(h, edge_index[1], colptr, col_perm)
+            edge_in = torch.cat(
+                [h_row, h_col, radial, edge_attr] if edge_attr is not None
+                else [h_row, h_col, radial], dim=1)
+            edge_feat = self.edge_mlp(edge_in)
+            if self.attention:
+                edge_feat = edge_feat * self.att_mlp(edge_feat)
 
         # --- global coord mean (site A collective) ----------------------
         coord_mean = ops.graph_mean_pool(coord, batch, b, ptr=ptr, counts=counts,
@@ -183,13 +201,14 @@ class EGCLVel(nn.Module):
             v_msg = v_msg * self.att_mlp_virtual(v_msg)
 
         # --- coord model (real), phi_x / phi_xv / phi_v ------------------
-        trans = coord_diff * self.coord_mlp_r(edge_feat)
-        if self.coords_agg == "mean":
-            agg = ops.segment_mean(trans, row, n, rowptr=rowptr)
-        elif self.coords_agg == "sum":
-            agg = ops.segment_sum(trans, row, n, rowptr=rowptr)
-        else:
-            raise ValueError(f"coords_agg {self.coords_agg}")
+        if not fuse:
+            trans = coord_diff * self.coord_mlp_r(edge_feat)
+            if self.coords_agg == "mean":
+                agg = ops.segment_mean(trans, row, n, rowptr=rowptr)
+            elif self.coords_agg == "sum":
+                agg = ops.segment_sum(trans, row, n, rowptr=rowptr)
+            else:
+                raise ValueError(f"coords_agg {self.coords_agg}")
         coord = coord + agg
         trans_v = (-vdiff * self.coord_mlp_r_virtual(v_msg)).mean(dim=1)
         coord = coord + trans_v
@@ -213,7 +232,8 @@ class EGCLVel(nn.Module):
         virtual_coord = virtual_coord + agg_vc
 
         # --- node model, phi_h ------------------------------------------
-        agg_e = ops.segment_mean(edge_feat, row, n, rowptr=rowptr)
+        if not fuse:
+            agg_e = ops.segment_mean(edge_feat, row, n, rowptr=rowptr)
         agg_v = v_msg.mean(dim=1)
         if node_attr is not None:
             node_in = torch.cat([h, agg_e, agg_v, node_attr], dim=1)

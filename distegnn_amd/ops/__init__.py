@@ -212,6 +212,100 @@ def gather_rows(x: torch.Tensor, idx: torch.Tensor,
     return _GatherRowsFn.apply(x, idx, segptr, perm, cb, ce, scp)
 
 
+def eager_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
+                     w1, b1, w2, b2, w3, b3, w3v, normalize, eps):
+    """Eager composition of the fused edge block (same math; used on CPU,
+    for numerics tests, and as the recompute path of the fused backward)."""
+    import torch.nn.functional as F
+
+    cd = (gather_rows(coord, row, rowptr)
+          - gather_rows(coord, col, colptr, col_perm))
+    radial = cd.pow(2).sum(1, keepdim=True)
+    if normalize:
+        cd = cd / (radial.sqrt().detach() + eps)
+    ein = torch.cat([gather_rows(h, row, rowptr),
+                     gather_rows(h, col, colptr, col_perm),
+                     radial.to(h.dtype), eattr.to(h.dtype)], dim=1)
+    t1 = F.silu(F.linear(ein, w1, b1))
+    msg = F.silu(F.linear(t1, w2, b2))
+    p = F.silu(F.linear(msg, w3, b3)) @ w3v.to(h.dtype)
+    trans = cd * p.unsqueeze(-1).float()
+    return msg, trans
+
+
+class _FusedEdgeBlockFn(torch.autograd.Function):
+    """Fused MFMA edge block (see csrc/fused_edge.hip) + CSR segment means.
+
+    Forward: one HIP kernel produces per-edge msg [M,64] bf16 and
+    trans [M,3] f32 (no [M,131]/[M,64] torch intermediates), then the CSR
+    segment-mean kernels aggregate both to nodes. Backward (v1): recompute
+    the eager composition under autograd — its gathers use the CSR
+    segment-sum backward, so no index_add scatters run."""
+
+    @staticmethod
+    def forward(ctx, h, coord, eattr, row, col, rowptr, colptr, col_perm,
+                w1, b1, w2, b2, w3, b3, w3v, normalize, eps):
+        ext = _require_ext("fused_edge_block")
+        msg, trans = ext.fused_edge_forward(
+            h, coord, eattr, row, col, w1.bfloat16(), b1, w2.bfloat16(), b2,
+            w3.bfloat16(), b3, w3v, bool(normalize), float(eps))
+        agg_msg = ext.segment_reduce_csr(msg, rowptr, True)
+        agg_trans = ext.segment_reduce_csr(trans, rowptr, True)
+        ctx.save_for_backward(h, coord, eattr, row, col, rowptr, colptr,
+                              col_perm, w1, b1, w2, b2, w3, b3, w3v)
+        ctx.normalize, ctx.eps = normalize, eps
+        return agg_msg, agg_trans
+
+    @staticmethod
+    def backward(ctx, dagg_msg, dagg_trans):
+        (h, coord, eattr, row, col, rowptr, colptr, col_perm,
+         w1, b1, w2, b2, w3, b3, w3v) = ctx.saved_tensors
+        deg = (rowptr[1:] - rowptr[:-1]).clamp(min=1)
+        dmsg = (dagg_msg / deg.unsqueeze(-1).to(dagg_msg.dtype)) \
+            .index_select(0, row)
+        dtrans = (dagg_trans / deg.unsqueeze(-1).to(dagg_trans.dtype)) \
+            .index_select(0, row)
+        with torch.enable_grad():
+            leaves = [t.detach().requires_grad_(True)
+                      for t in (h, coord, w1, b1, w2, b2, w3, b3, w3v)]
+            hh, cc, lw1, lb1, lw2, lb2, lw3, lb3, lw3v = leaves
+            msg, trans = eager_edge_block(
+                hh, cc, eattr, row, col, rowptr, colptr, col_perm,
+                lw1.to(h.dtype), lb1, lw2.to(h.dtype), lb2, lw3.to(h.dtype),
+                lb3, lw3v, ctx.normalize, ctx.eps)
+            grads = torch.autograd.grad(
+                (msg, trans), leaves,
+                grad_outputs=(dmsg.to(msg.dtype), dtrans.to(trans.dtype)),
+                allow_unused=True)
+        gh, gc, gw1, gb1, gw2, gb2, gw3, gb3, gw3v = grads
+        return (gh, gc, None, None, None, None, None, None,
+                gw1, gb1, gw2, gb2, gw3, gb3, gw3v, None, None)
+
+
+def fused_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
+                     w1, b1, w2, b2, w3, b3, w3v, normalize, eps):
+    """Dispatch: HIP fused kernel on GPU bf16 H=64, eager otherwise.
+
+    Returns (agg_msg [N,64], agg_trans [N,3]) — per-node MEANS of the edge
+    messages and coordinate translations."""
+    usable = (h.is_cuda and h.dtype == torch.bfloat16 and h.size(1) == 64
+              and eattr is not None and eattr.size(1) == 2
+              and rowptr is not None and colptr is not None
+              and col_perm is not None and hip_ext() is not None
+              and os.environ.get("DISTEGNN_DISABLE_FUSED") != "1")
+    if usable:
+        return _FusedEdgeBlockFn.apply(
+            h, coord, eattr, row, col, rowptr, colptr, col_perm,
+            w1, b1, w2, b2, w3, b3, w3v, normalize, eps)
+    msg, trans = eager_edge_block(
+        h, coord, eattr, row, col, rowptr, colptr, col_perm,
+        w1.to(h.dtype), b1, w2.to(h.dtype), b2, w3.to(h.dtype), b3, w3v,
+        normalize, eps)
+    n = coord.size(0)
+    return (segment_mean(msg, row, n, rowptr=rowptr),
+            segment_mean(trans, row, n, rowptr=rowptr))
+
+
 def radius_graph(pos: torch.Tensor, r: float, loop: bool = False) -> torch.Tensor:
     """Directed radius graph, row-sorted. GPU: HIP cell-list kernel."""
     if pos.is_cuda and r is not None and r >= 0:
@@ -223,5 +317,6 @@ def radius_graph(pos: torch.Tensor, r: float, loop: bool = False) -> torch.Tenso
 
 __all__ = [
     "segment_sum", "segment_mean", "graph_sum_pool", "graph_mean_pool",
-    "gather_rows", "radius_graph", "hip_ext", "reference",
+    "gather_rows", "fused_edge_block", "eager_edge_block", "radius_graph",
+    "hip_ext", "reference",
 ]

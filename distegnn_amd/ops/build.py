@@ -21,6 +21,7 @@ SOURCES = [
     os.path.join(_CSRC, "ext.cpp"),
     os.path.join(_CSRC, "segment_reduce.hip"),
     os.path.join(_CSRC, "radius.hip"),
+    os.path.join(_CSRC, "fused_edge.hip"),
 ]
 
 

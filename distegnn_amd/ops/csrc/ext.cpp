@@ -9,6 +9,12 @@ torch::Tensor segment_reduce_chunked(torch::Tensor data, torch::Tensor rowptr,
                                      torch::Tensor seg_chunk_ptr, bool mean);
 std::tuple<torch::Tensor, torch::Tensor> radius_graph_gpu(torch::Tensor pos,
                                                           double r);
+std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
+    torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
+    torch::Tensor row, torch::Tensor col, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    torch::Tensor w3v, bool normalize, double eps);
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
 
 torch::Tensor radius_graph(torch::Tensor pos, double r) {
   return std::get<0>(radius_graph_gpu(pos, r));
@@ -28,4 +34,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("r"));
   m.def("radius_graph", &radius_graph, "cell-list radius graph edge_index",
         py::arg("pos"), py::arg("r"));
+  m.def("fused_edge_forward", &fused_edge_forward,
+        "fused MFMA edge block: gather + phi_e MLP + phi_x head + trans",
+        py::arg("h"), py::arg("coord"), py::arg("eattr"), py::arg("row"),
+        py::arg("col"), py::arg("w1"), py::arg("b1"), py::arg("w2"),
+        py::arg("b2"), py::arg("w3"), py::arg("b3"), py::arg("w3v"),
+        py::arg("normalize"), py::arg("eps"));
+  m.def("mfma_probe", &mfma_probe,
+        "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
+        py::arg("a"), py::arg("bt"));
 }

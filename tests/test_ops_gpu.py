@@ -74,7 +74,7 @@ def test_chunked_pool_matches(f):
                                     torch.tensor([30000, 20000]))
     x = torch.randn(n, f)
     want = R.graph_mean_pool(x, batch, b)
-    chunk = 2048
+    chunk = 256
     cb, ce, scp = [], [], [0]
     for i in range(b):
         for k in range(int(ptr[i]), int(ptr[i + 1]), chunk):
