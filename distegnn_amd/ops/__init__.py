@@ -218,6 +218,9 @@ def eager_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
     for numerics tests, and as the recompute path of the fused backward)."""
     import torch.nn.functional as F
 
+    dt = h.dtype
+    w1, b1, w2, b2, w3, b3, w3v = (t.to(dt)
+                                   for t in (w1, b1, w2, b2, w3, b3, w3v))
     cd = (gather_rows(coord, row, rowptr)
           - gather_rows(coord, col, colptr, col_perm))
     radial = cd.pow(2).sum(1, keepdim=True)
@@ -228,7 +231,7 @@ def eager_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
                      radial.to(h.dtype), eattr.to(h.dtype)], dim=1)
     t1 = F.silu(F.linear(ein, w1, b1))
     msg = F.silu(F.linear(t1, w2, b2))
-    p = F.silu(F.linear(msg, w3, b3)) @ w3v.to(h.dtype)
+    p = F.silu(F.linear(msg, w3, b3)) @ w3v
     trans = cd * p.unsqueeze(-1).float()
     return msg, trans
 
@@ -271,8 +274,8 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             hh, cc, lw1, lb1, lw2, lb2, lw3, lb3, lw3v = leaves
             msg, trans = eager_edge_block(
                 hh, cc, eattr, row, col, rowptr, colptr, col_perm,
-                lw1.to(h.dtype), lb1, lw2.to(h.dtype), lb2, lw3.to(h.dtype),
-                lb3, lw3v, ctx.normalize, ctx.eps)
+                lw1, lb1, lw2, lb2, lw3, lb3, lw3v,
+                ctx.normalize, ctx.eps)
             grads = torch.autograd.grad(
                 (msg, trans), leaves,
                 grad_outputs=(dmsg.to(msg.dtype), dtrans.to(trans.dtype)),
@@ -299,8 +302,7 @@ def fused_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
             w1, b1, w2, b2, w3, b3, w3v, normalize, eps)
     msg, trans = eager_edge_block(
         h, coord, eattr, row, col, rowptr, colptr, col_perm,
-        w1.to(h.dtype), b1, w2.to(h.dtype), b2, w3.to(h.dtype), b3, w3v,
-        normalize, eps)
+        w1, b1, w2, b2, w3, b3, w3v, normalize, eps)
     n = coord.size(0)
     return (segment_mean(msg, row, n, rowptr=rowptr),
             segment_mean(trans, row, n, rowptr=rowptr))
