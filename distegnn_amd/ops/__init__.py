@@ -264,6 +264,35 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
         (h, coord, eattr, row, col, rowptr, colptr, col_perm,
          w1, b1, w2, b2, w3, b3, w3v) = ctx.saved_tensors
         deg = (rowptr[1:] - rowptr[:-1]).clamp(min=1)
+        ext = _load_extension()
+        if ext is not None and os.environ.get(
+                "DISTEGNN_FUSED_BWD_RECOMPUTE") != "1":
+            # fused backward kernel: in-LDS recompute + per-edge grads
+            dmsg_n = (dagg_msg / deg.unsqueeze(-1).to(dagg_msg.dtype)) \
+                .to(torch.bfloat16).contiguous()
+            dtrans_n = (dagg_trans
+                        / deg.unsqueeze(-1).to(dagg_trans.dtype)).contiguous()
+            (ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd,
+             gw3v) = ext.fused_edge_backward(
+                h, coord, eattr, row, col, dmsg_n, dtrans_n,
+                w1.bfloat16(), b1, w2.bfloat16(), b2, w3.bfloat16(), b3,
+                w3v, bool(ctx.normalize), float(ctx.eps))
+            gh = (ext.segment_reduce_csr(dhr, rowptr, False)
+                  + ext.segment_reduce_csr(dhc.index_select(0, col_perm),
+                                           colptr, False))
+            gc = (ext.segment_reduce_csr(dcd, rowptr, False)
+                  - ext.segment_reduce_csr(dcd.index_select(0, col_perm),
+                                           colptr, False))
+            k_in = w1.size(1)
+            gw1 = torch.mm(dz1.t(), ein)[:, :k_in].float()
+            gw2 = torch.mm(dz2.t(), t1).float()
+            gw3 = torch.mm(dz3.t(), msg).float()
+            gb1 = dz1.sum(0).float()
+            gb2 = dz2.sum(0).float()
+            gb3 = dz3.sum(0).float()
+            return (gh.to(h.dtype), gc, None, None, None, None, None, None,
+                    gw1, gb1, gw2, gb2, gw3, gb3, gw3v, None, None)
+        # fallback: recompute the eager composition under autograd
         dmsg = (dagg_msg / deg.unsqueeze(-1).to(dagg_msg.dtype)) \
             .index_select(0, row)
         dtrans = (dagg_trans / deg.unsqueeze(-1).to(dagg_trans.dtype)) \

@@ -22,6 +22,7 @@ SOURCES = [
     os.path.join(_CSRC, "segment_reduce.hip"),
     os.path.join(_CSRC, "radius.hip"),
     os.path.join(_CSRC, "fused_edge.hip"),
+    os.path.join(_CSRC, "fused_edge_bwd.hip"),
 ]
 
 

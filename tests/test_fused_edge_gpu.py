@@ -104,6 +104,41 @@ def test_fused_block_autograd_vs_eager():
         assert torch.allclose(a, b, atol=0.3, rtol=0.1)
 
 
+def test_fused_backward_kernel_vs_recompute():
+    """The fused backward kernel == the autograd recompute backward."""
+    import os
+
+    bt = make_graph(n=2500)
+    params = make_params(3)
+    h0 = torch.randn(bt.num_nodes, 64, device=dev()) * 0.5
+
+    def run(mode):
+        if mode == "recompute":
+            os.environ["DISTEGNN_FUSED_BWD_RECOMPUTE"] = "1"
+        else:
+            os.environ.pop("DISTEGNN_FUSED_BWD_RECOMPUTE", None)
+        ps = [p.detach().clone().requires_grad_(True) for p in params]
+        h = h0.detach().clone().bfloat16().requires_grad_(True)
+        coord = bt.pos.detach().clone().requires_grad_(True)
+        agg_msg, agg_trans = ops.fused_edge_block(
+            h, coord, bt.edge_attr, bt.edge_index[0], bt.edge_index[1],
+            bt.rowptr, bt.colptr, bt.col_perm, *ps, True, 1e-8)
+        (agg_msg.float().pow(2).sum() + agg_trans.pow(2).sum()).backward()
+        os.environ.pop("DISTEGNN_FUSED_BWD_RECOMPUTE", None)
+        return h.grad.float(), coord.grad, [p.grad for p in ps]
+
+    gh_k, gc_k, gp_k = run("kernel")
+    gh_r, gc_r, gp_r = run("recompute")
+    assert torch.allclose(gh_k, gh_r, atol=0.05, rtol=0.1), \
+        (gh_k - gh_r).abs().max()
+    assert torch.allclose(gc_k, gc_r, atol=0.05, rtol=0.1), \
+        (gc_k - gc_r).abs().max()
+    names = ["w1", "b1", "w2", "b2", "w3", "b3", "w3v"]
+    for nm, a, b in zip(names, gp_k, gp_r):
+        assert torch.allclose(a, b, atol=0.25, rtol=0.1), \
+            (nm, (a - b).abs().max())
+
+
 def test_fused_tail_tile():
     """M not divisible by 64 handled (tail edges)."""
     bt = make_graph(n=500)
