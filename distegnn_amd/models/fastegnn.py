@@ -45,7 +45,12 @@ import torch
 from torch import nn
 
 from .. import ops
+from ..ops.linear import SplitKLinear
 from ..parallel import comm
+
+# nn.Linear drop-in with split-K weight gradients for the [N]/[N*C]-row
+# activations (state-dict compatible; see ops/linear.py)
+Linear = SplitKLinear
 
 
 class EGCLVel(nn.Module):
@@ -69,25 +74,25 @@ class EGCLVel(nn.Module):
         edge_coords_nf = 1
 
         self.edge_mlp = nn.Sequential(  # phi_e
-            nn.Linear(2 * node_feat_nf + edge_coords_nf + edge_attr_nf, hidden_nf),
+            Linear(2 * node_feat_nf + edge_coords_nf + edge_attr_nf, hidden_nf),
             act_fn,
-            nn.Linear(hidden_nf, hidden_nf),
+            Linear(hidden_nf, hidden_nf),
             act_fn,
         )
         self.edge_mlp_virtual = nn.Sequential(  # phi_ev (no edge feat)
-            nn.Linear(2 * node_feat_nf + edge_coords_nf + virtual_channels, hidden_nf),
+            Linear(2 * node_feat_nf + edge_coords_nf + virtual_channels, hidden_nf),
             act_fn,
-            nn.Linear(hidden_nf, hidden_nf),
+            Linear(hidden_nf, hidden_nf),
             act_fn,
         )
         if attention:
-            self.att_mlp = nn.Sequential(nn.Linear(hidden_nf, 1), nn.Sigmoid())
-            self.att_mlp_virtual = nn.Sequential(nn.Linear(hidden_nf, 1), nn.Sigmoid())
+            self.att_mlp = nn.Sequential(Linear(hidden_nf, 1), nn.Sigmoid())
+            self.att_mlp_virtual = nn.Sequential(Linear(hidden_nf, 1), nn.Sigmoid())
 
         def coord_head():
-            last = nn.Linear(hidden_nf, 1, bias=False)
+            last = Linear(hidden_nf, 1, bias=False)
             nn.init.xavier_uniform_(last.weight, gain=0.001)
-            mods = [nn.Linear(hidden_nf, hidden_nf), act_fn, last]
+            mods = [Linear(hidden_nf, hidden_nf), act_fn, last]
             if tanh:
                 mods.append(nn.Tanh())
             return nn.Sequential(*mods)
@@ -96,22 +101,22 @@ class EGCLVel(nn.Module):
         self.coord_mlp_r_virtual = coord_head()    # phi_xv
         self.coord_mlp_v_virtual = coord_head()    # phi_X
         self.coord_mlp_vel = nn.Sequential(        # phi_v
-            nn.Linear(node_feat_nf, hidden_nf), act_fn, nn.Linear(hidden_nf, 1)
+            Linear(node_feat_nf, hidden_nf), act_fn, Linear(hidden_nf, 1)
         )
         self.gravity = gravity
         if gravity is not None:
             self.gravity_mlp = nn.Sequential(
-                nn.Linear(node_feat_nf, hidden_nf), act_fn, nn.Linear(hidden_nf, 1)
+                Linear(node_feat_nf, hidden_nf), act_fn, Linear(hidden_nf, 1)
             )
         self.node_mlp = nn.Sequential(  # phi_h
-            nn.Linear(3 * hidden_nf + node_attr_nf, hidden_nf),
+            Linear(3 * hidden_nf + node_attr_nf, hidden_nf),
             act_fn,
-            nn.Linear(hidden_nf, node_feat_out_nf),
+            Linear(hidden_nf, node_feat_out_nf),
         )
         self.node_mlp_virtual = nn.Sequential(  # phi_hv
-            nn.Linear(2 * hidden_nf, hidden_nf),
+            Linear(2 * hidden_nf, hidden_nf),
             act_fn,
-            nn.Linear(hidden_nf, node_feat_out_nf),
+            Linear(hidden_nf, node_feat_out_nf),
         )
 
     # --- geometry -------------------------------------------------------
@@ -270,7 +275,7 @@ class FastEGNN(nn.Module):
         # [1, H, C]: reference-compatible parameter shape (FastEGNN.py:288)
         self.virtual_node_feat = nn.Parameter(
             torch.randn(1, hidden_nf, virtual_channels))
-        self.embedding_in = nn.Linear(node_feat_nf, hidden_nf)
+        self.embedding_in = Linear(node_feat_nf, hidden_nf)
         if gravity is not None:
             gravity = torch.tensor(gravity)
         for i in range(n_layers):

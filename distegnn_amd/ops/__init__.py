@@ -283,10 +283,12 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             gc = (ext.segment_reduce_csr(dcd, rowptr, False)
                   - ext.segment_reduce_csr(dcd.index_select(0, col_perm),
                                            colptr, False))
+            from .linear import chunked_wgrad
+
             k_in = w1.size(1)
-            gw1 = torch.mm(dz1.t(), ein)[:, :k_in].float()
-            gw2 = torch.mm(dz2.t(), t1).float()
-            gw3 = torch.mm(dz3.t(), msg).float()
+            gw1 = chunked_wgrad(dz1, ein)[:, :k_in].float()
+            gw2 = chunked_wgrad(dz2, t1).float()
+            gw3 = chunked_wgrad(dz3, msg).float()
             gb1 = dz1.sum(0).float()
             gb2 = dz2.sum(0).float()
             gb3 = dz3.sum(0).float()

@@ -1,0 +1,76 @@
+"""Split-K aware Linear for tall-skinny activations.
+
+hipBLASLt's heuristic picks a non-split-K kernel for weight-gradient GEMMs
+of shape [64..144] x [M ~ 10^6] (K = M deep): ~6 workgroups on 256 CUs,
+measured 2.9 ms where the roofline is ~0.1 ms. ``chunked_wgrad`` performs
+the split-K manually as a batched GEMM over row chunks + a partial sum
+(measured 0.23 ms on [1.65M,64]^T @ [1.65M,144]).
+
+``SplitKLinear`` is a drop-in ``nn.Linear`` (same state-dict keys) whose
+backward uses the chunked weight gradient for large inputs.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+_CHUNKS = 64
+_MIN_ROWS = 1 << 16
+
+
+def chunked_wgrad(g: torch.Tensor, x: torch.Tensor,
+                  nc: int = _CHUNKS) -> torch.Tensor:
+    """g [M, O], x [M, I] -> g^T @ x [O, I] via nc-way split-K bmm."""
+    m = g.size(0)
+    mc = m // nc
+    if mc == 0:
+        return torch.mm(g.t(), x)
+    main = nc * mc
+    out = torch.bmm(g[:main].view(nc, mc, -1).transpose(1, 2),
+                    x[:main].view(nc, mc, -1)).sum(0)
+    if main < m:
+        out = out + torch.mm(g[main:].t(), x[main:])
+    return out
+
+
+class _SplitKLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return F.linear(x, w, b)
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w = ctx.saved_tensors
+        g = g.contiguous()
+        gx = g @ w.to(g.dtype)
+        gw = chunked_wgrad(g, x.to(g.dtype)).to(w.dtype)
+        gb = g.sum(0).to(w.dtype) if ctx.has_bias else None
+        return gx, gw, gb
+
+
+class SplitKLinear(nn.Linear):
+    """nn.Linear with manual split-K weight gradients for tall inputs."""
+
+    def forward(self, x):
+        if (x.is_cuda and x.dim() >= 2
+                and x.numel() // x.size(-1) >= _MIN_ROWS):
+            shape = x.shape
+            x2 = x.reshape(-1, shape[-1])
+            w = self.weight
+            b = self.bias
+            if torch.is_autocast_enabled():
+                dt = torch.get_autocast_dtype("cuda")
+                x2 = x2.to(dt)
+                w = w.to(dt)
+                b = b.to(dt) if b is not None else None
+            elif x2.dtype != w.dtype and x2.dtype in (torch.bfloat16,
+                                                      torch.float16):
+                w = w.to(x2.dtype)
+                b = b.to(x2.dtype) if b is not None else None
+            out = _SplitKLinearFn.apply(x2.contiguous(), w, b)
+            return out.reshape(*shape[:-1], out.size(-1))
+        return super().forward(x)
