@@ -23,6 +23,7 @@ SOURCES = [
     os.path.join(_CSRC, "radius.hip"),
     os.path.join(_CSRC, "fused_edge.hip"),
     os.path.join(_CSRC, "fused_edge_bwd.hip"),
+    os.path.join(_CSRC, "wgrad.hip"),
 ]
 
 

@@ -15,6 +15,7 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
     torch::Tensor w3v, bool normalize, double eps);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
+torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
 std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
@@ -53,6 +54,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
         py::arg("w3"), py::arg("b3"), py::arg("w3v"), py::arg("normalize"),
         py::arg("eps"));
+  m.def("wgrad_splitk", &wgrad_splitk_launch,
+        "split-K MFMA weight gradient: g^T @ x for tall activations",
+        py::arg("g"), py::arg("x"));
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
         py::arg("a"), py::arg("bt"));

@@ -1,0 +1,163 @@
+// Split-K weight-gradient kernel: dW = g^T @ x for tall-skinny activations.
+//
+// Shapes: g [M, 64] bf16 (layer-output grad), x [M, I<=208] bf16
+// (activation), M ~ 10^5..10^6. hipBLASLt's heuristic picks a non-split-K
+// kernel here (6 workgroups on 256 CUs, 2.9 ms measured); torch's bmm
+// split-K workaround costs 2.6 ms of HOST time per call re-running the
+// batched-GEMM heuristic. This kernel owns the shape: each block reduces a
+// row chunk with mfma_f32_16x16x32_bf16 over LDS-transposed sub-tiles and
+// writes an fp32 partial; the [nchunk, 64, I] partials are summed by a
+// tiny torch reduction.
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int O_DIM = 64;       // rows of dW (layer width) — fixed
+constexpr int THREADS = 256;    // 4 waves
+constexpr int E_STEP = 32;      // MFMA K per step
+constexpr int CHUNK = 2048;     // rows per block (split-K granularity)
+constexpr int T_STRIDE = 40;    // LDS e-stride (bank-conflict pad)
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// NTW = n-tiles (of 16 cols of x) per wave; ITILES total = waves used * NTW
+// mapping: wave w covers n-tiles [w*NTW, w*NTW+NTW)
+template <int NTW>
+__global__ __launch_bounds__(THREADS) void wgrad_splitk(
+    const bf16* __restrict__ g,  // [M, 64]
+    const bf16* __restrict__ x,  // [M, I]
+    float* __restrict__ part,    // [nchunk, 64, IP] (IP = 16*ceil(I/16))
+    long m, int i_dim, int ip) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // LDS: gT [64][T_STRIDE] bf16, xT [ip<=208][T_STRIDE] bf16
+  __bf16* gT = reinterpret_cast<__bf16*>(smem);
+  __bf16* xT = gT + O_DIM * T_STRIDE;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int itiles = ip / 16;
+  const int ntw_used = (itiles + 3) / 4;  // n-tiles this wave actually owns
+
+  long c0 = (long)blockIdx.x * CHUNK;
+  long c1 = c0 + CHUNK < m ? c0 + CHUNK : m;
+
+  f32x4 acc[4][NTW] = {};
+
+  for (long e0 = c0; e0 < c1; e0 += E_STEP) {
+    int ne = (int)((c1 - e0 < E_STEP) ? (c1 - e0) : (long)E_STEP);
+    __syncthreads();
+    // stage g[e0..e0+31][0..63] transposed -> gT[o][e]
+    for (int idx = tid; idx < E_STEP * (O_DIM / 8); idx += THREADS) {
+      int e = idx / (O_DIM / 8);
+      int o8 = (idx % (O_DIM / 8)) * 8;
+      bf16x8 v = {};
+      if (e < ne) v = *reinterpret_cast<const bf16x8*>(g + (e0 + e) * O_DIM + o8);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) gT[(o8 + u) * T_STRIDE + e] = v[u];
+    }
+    // stage x[e][i] transposed -> xT[i][e] (i padded with zeros)
+    for (int idx = tid; idx < E_STEP * (ip / 8); idx += THREADS) {
+      int e = idx / (ip / 8);
+      int i8 = (idx % (ip / 8)) * 8;
+      if (e < ne && i8 + 8 <= i_dim) {
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(x + (e0 + e) * i_dim + i8);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) xT[(i8 + u) * T_STRIDE + e] = v[u];
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          int i = i8 + u;
+          __bf16 val = (__bf16)0.f;
+          if (e < ne && i < i_dim) val = ((const __bf16*)x)[(e0 + e) * i_dim + i];
+          xT[i * T_STRIDE + e] = val;
+        }
+      }
+    }
+    __syncthreads();
+    // D[o][i] += gT[o][e] * xT[i][e] — A rows = o, B cols = i, K = e
+    int kb = (lane >> 4) * 8;
+#pragma unroll
+    for (int nt = 0; nt < NTW; ++nt) {
+      int icol = (wave * NTW + nt) * 16 + (lane & 15);
+      if (wave * NTW + nt >= itiles) break;
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(&xT[icol * T_STRIDE + kb]);
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &gT[(mt * 16 + (lane & 15)) * T_STRIDE + kb]);
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt][nt],
+                                                              0, 0, 0);
+      }
+    }
+  }
+
+  // write partial [64][ip]: C layout col=l&15(+16*(w*NTW+nt)), row=(l>>4)*4+r
+  float* out = part + (long)blockIdx.x * O_DIM * ip;
+#pragma unroll
+  for (int nt = 0; nt < NTW; ++nt) {
+    if (wave * NTW + nt >= itiles) break;
+    int i = (wave * NTW + nt) * 16 + (lane & 15);
+#pragma unroll
+    for (int mt = 0; mt < 4; ++mt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int o = mt * 16 + (lane >> 4) * 4 + r;
+        out[o * ip + i] = acc[mt][nt][r];
+      }
+    }
+  }
+  (void)ntw_used;
+}
+
+}  // namespace
+
+torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
+  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16,
+              "g must be CUDA bf16");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(g.size(1) == O_DIM, "wgrad kernel requires out width 64");
+  auto gc = g.contiguous();
+  auto xc = x.contiguous();
+  long m = gc.size(0);
+  int i_dim = (int)xc.size(1);
+  TORCH_CHECK(i_dim <= 208, "wgrad kernel supports I<=208");
+  int ip = ((i_dim + 15) / 16) * 16;
+  long nchunk = (m + CHUNK - 1) / CHUNK;
+  auto part = torch::empty({nchunk, (long)O_DIM, (long)ip},
+                           g.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  int smem = (O_DIM + ip) * T_STRIDE * 2;
+  int itiles = ip / 16;
+  int ntw = (itiles + 3) / 4;
+  const bf16* gp = reinterpret_cast<const bf16*>(gc.data_ptr());
+  const bf16* xp = reinterpret_cast<const bf16*>(xc.data_ptr());
+  float* pp = part.data_ptr<float>();
+  switch (ntw) {
+    case 1:
+      wgrad_splitk<1><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
+                                                              i_dim, ip);
+      break;
+    case 2:
+      wgrad_splitk<2><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
+                                                              i_dim, ip);
+      break;
+    case 3:
+      wgrad_splitk<3><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
+                                                              i_dim, ip);
+      break;
+    case 4:
+      wgrad_splitk<4><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
+                                                              i_dim, ip);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported I for wgrad kernel");
+  }
+  auto dw = part.sum(0);  // [64, ip]
+  return dw.narrow(1, 0, i_dim);
+}

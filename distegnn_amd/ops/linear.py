@@ -22,7 +22,18 @@ _MIN_ROWS = 1 << 16
 
 def chunked_wgrad(g: torch.Tensor, x: torch.Tensor,
                   nc: int = _CHUNKS) -> torch.Tensor:
-    """g [M, O], x [M, I] -> g^T @ x [O, I] via nc-way split-K bmm."""
+    """g [M, O], x [M, I] -> g^T @ x [O, I], split-K.
+
+    Primary path: the hand-written MFMA split-K kernel (csrc/wgrad.hip) for
+    the standard O=64 bf16 case. (A torch.bmm split-K was tried first and
+    costs 2.6 ms of HOST time per call in hipBLASLt's batched heuristic.)"""
+    from . import hip_ext
+
+    ext = hip_ext()
+    if (ext is not None and g.is_cuda and g.dtype == torch.bfloat16
+            and g.size(1) == 64 and x.size(1) <= 208
+            and x.dtype == torch.bfloat16):
+        return ext.wgrad_splitk(g, x)
     m = g.size(0)
     mc = m // nc
     if mc == 0:

@@ -153,3 +153,15 @@ def test_fused_tail_tile():
     assert msg.shape == (m, 64) and trans.shape == (m, 3)
     assert torch.isfinite(msg.float()).all()
     assert torch.isfinite(trans).all()
+
+
+@pytest.mark.parametrize("i_dim", [64, 72, 134, 144, 194])
+def test_wgrad_splitk_matches_mm(i_dim):
+    torch.manual_seed(0)
+    m = 200_000 + 37  # non-divisible tail
+    g = (torch.randn(m, 64, device=dev()) * 0.1).bfloat16()
+    x = (torch.randn(m, i_dim, device=dev()) * 0.1).bfloat16()
+    want = torch.mm(g.t().float(), x.float())
+    got = ops.hip_ext().wgrad_splitk(g, x)
+    rel = (got - want).norm() / want.norm().clamp(min=1e-9)
+    assert rel < 2e-2, rel.item()
