@@ -34,6 +34,10 @@ def chunked_wgrad(g: torch.Tensor, x: torch.Tensor,
             and g.size(1) == 64 and x.size(1) <= 208
             and x.dtype == torch.bfloat16):
         return ext.wgrad_splitk(g, x)
+    if g.size(1) < 16:
+        # thin-output wgrad ([M,1] head grads): hipBLASLt's skinny kernel is
+        # fine here, and the bmm split-K costs ~2.6 ms HOST per call.
+        return torch.mm(g.t(), x)
     m = g.size(0)
     mc = m // nc
     if mc == 0:
