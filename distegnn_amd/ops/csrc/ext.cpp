@@ -16,6 +16,8 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     torch::Tensor w3v, bool normalize, double eps);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
+torch::Tensor tall_linear(torch::Tensor x, torch::Tensor bmat,
+                          c10::optional<torch::Tensor> bias, int64_t act);
 std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
@@ -57,6 +59,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_splitk", &wgrad_splitk_launch,
         "split-K MFMA weight gradient: g^T @ x for tall activations",
         py::arg("g"), py::arg("x"));
+  m.def("tall_linear", &tall_linear,
+        "tall-skinny MFMA linear: act(x @ B^T + bias)",
+        py::arg("x"), py::arg("bmat"), py::arg("bias") = c10::nullopt,
+        py::arg("act") = 0);
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
         py::arg("a"), py::arg("bt"));

@@ -50,18 +50,36 @@ def chunked_wgrad(g: torch.Tensor, x: torch.Tensor,
     return out
 
 
+def _ext_ok(x, w):
+    from . import hip_ext
+
+    return (hip_ext() is not None and x.is_cuda
+            and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16
+            and w.size(0) <= 208 and w.size(1) <= 208)
+
+
 class _SplitKLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
+        if _ext_ok(x, w):
+            from . import hip_ext
+
+            return hip_ext().tall_linear(x, w, b, 0)
         return F.linear(x, w, b)
 
     @staticmethod
     def backward(ctx, g):
         x, w = ctx.saved_tensors
         g = g.contiguous()
-        gx = g @ w.to(g.dtype)
+        wt = w.to(g.dtype)
+        if _ext_ok(g, wt.t()):
+            from . import hip_ext
+
+            gx = hip_ext().tall_linear(g, wt.t().contiguous(), None, 0)
+        else:
+            gx = g @ wt
         gw = chunked_wgrad(g, x.to(g.dtype)).to(w.dtype)
         gb = g.sum(0).to(w.dtype) if ctx.has_bias else None
         return gx, gw, gb

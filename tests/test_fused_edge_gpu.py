@@ -165,3 +165,21 @@ def test_wgrad_splitk_matches_mm(i_dim):
     got = ops.hip_ext().wgrad_splitk(g, x)
     rel = (got - want).norm() / want.norm().clamp(min=1e-9)
     assert rel < 2e-2, rel.item()
+
+
+@pytest.mark.parametrize("k,o,bias,act", [(134, 64, True, 0), (64, 64, True, 1),
+                                          (64, 1, False, 0), (194, 64, True, 0),
+                                          (64, 134, False, 0)])
+def test_tall_linear_matches_torch(k, o, bias, act):
+    torch.manual_seed(1)
+    m = 70_000 + 11
+    x = (torch.randn(m, k, device=dev()) * 0.3).bfloat16()
+    w = (torch.randn(o, k, device=dev()) * 0.2).bfloat16()
+    b = (torch.randn(o, device=dev()) * 0.1) if bias else None
+    got = ops.hip_ext().tall_linear(x, w, b, act)
+    want = torch.nn.functional.linear(x.float(), w.float(),
+                                      b.float() if bias else None)
+    if act == 1:
+        want = torch.nn.functional.silu(want)
+    rel = (got.float() - want).norm() / want.norm().clamp(min=1e-9)
+    assert rel < 2e-2, rel.item()
