@@ -19,7 +19,7 @@ namespace {
 constexpr int O_DIM = 64;       // rows of dW (layer width) — fixed
 constexpr int THREADS = 256;    // 4 waves
 constexpr int E_STEP = 32;      // MFMA K per step
-constexpr int CHUNK = 2048;     // rows per block (split-K granularity)
+// split-K granularity: chunk chosen at launch for ~1024 blocks
 constexpr int T_STRIDE = 40;    // LDS e-stride (bank-conflict pad)
 
 using bf16 = __hip_bfloat16;
@@ -33,7 +33,7 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
     const bf16* __restrict__ g,  // [M, 64]
     const bf16* __restrict__ x,  // [M, I]
     float* __restrict__ part,    // [nchunk, 64, IP] (IP = 16*ceil(I/16))
-    long m, int i_dim, int ip) {
+    long m, int i_dim, int ip, long chunk) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // LDS: gT [64][T_STRIDE] bf16, xT [ip<=208][T_STRIDE] bf16
   __bf16* gT = reinterpret_cast<__bf16*>(smem);
@@ -44,8 +44,8 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
   const int itiles = ip / 16;
   const int ntw_used = (itiles + 3) / 4;  // n-tiles this wave actually owns
 
-  long c0 = (long)blockIdx.x * CHUNK;
-  long c1 = c0 + CHUNK < m ? c0 + CHUNK : m;
+  long c0 = (long)blockIdx.x * chunk;
+  long c1 = c0 + chunk < m ? c0 + chunk : m;
 
   f32x4 acc[4][NTW] = {};
 
@@ -128,7 +128,9 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   int i_dim = (int)xc.size(1);
   TORCH_CHECK(i_dim <= 208, "wgrad kernel supports I<=208");
   int ip = ((i_dim + 15) / 16) * 16;
-  long nchunk = (m + CHUNK - 1) / CHUNK;
+  long chunk = ((m + 1023) / 1024 + 31) / 32 * 32;  // ~1024 blocks
+  if (chunk < 64) chunk = 64;
+  long nchunk = (m + chunk - 1) / chunk;
   auto part = torch::empty({nchunk, (long)O_DIM, (long)ip},
                            g.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
@@ -141,19 +143,19 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   switch (ntw) {
     case 1:
       wgrad_splitk<1><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip);
+                                                              i_dim, ip, chunk);
       break;
     case 2:
       wgrad_splitk<2><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip);
+                                                              i_dim, ip, chunk);
       break;
     case 3:
       wgrad_splitk<3><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip);
+                                                              i_dim, ip, chunk);
       break;
     case 4:
       wgrad_splitk<4><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip);
+                                                              i_dim, ip, chunk);
       break;
     default:
       TORCH_CHECK(false, "unsupported I for wgrad kernel");
