@@ -34,9 +34,11 @@ def chunked_wgrad(g: torch.Tensor, x: torch.Tensor,
             and g.size(1) == 64 and x.size(1) <= 208
             and x.dtype == torch.bfloat16):
         return ext.wgrad_splitk(g, x)
+    if g.size(1) == 1:
+        # head wgrad [1,M]x[M,I]: a weighted column sum — the library's
+        # MT64x16x512 pick costs ~0.5 ms; mul+reduce is ~0.1 ms
+        return (g * x).sum(0, keepdim=True, dtype=torch.float32).to(g.dtype)
     if g.size(1) < 16:
-        # thin-output wgrad ([M,1] head grads): hipBLASLt's skinny kernel is
-        # fine here, and the bmm split-K costs ~2.6 ms HOST per call.
         return torch.mm(g.t(), x)
     m = g.size(0)
     mc = m // nc
