@@ -8,18 +8,15 @@
 //   s3     = SiLU(msg   @ W3^T + b3)            [H]   (phi_x hidden)
 //   p_e    = s3 . w3                            scalar (phi_x head)
 //   trans  = d_e * p_e                          [3]
-// and writes msg [M,H] bf16 + trans [M,3] f32 (aggregated to nodes by the
-// CSR segment-mean kernel). This replaces, per layer: 2 gathers, a [M,131]
-// concat, 3 hipBLASLt GEMMs, 3 SiLU kernels and the coord_diff/radial
-// elementwise ops (reference models/FastEGNN.py:144-150, 166-173, 237-246)
-// — all [M,.] intermediates except the two outputs stay in LDS/registers.
+// writing msg [M,H] bf16 + trans [M,3] f32 (aggregated by the CSR
+// segment-mean kernels). All [M,.] intermediates stay in LDS/registers.
 //
-// Mapping: 256 threads = 4 waves per block; each block owns a tile of 64
-// row-sorted edges; each wave computes a 16-edge x 64-feature MFMA tile
-// (mfma_f32_16x16x32_bf16, fp32 accumulation). Inputs are gathered into an
-// LDS tile [64][K_STRIDE] (bank-conflict-padded); W1/W2/W3 are staged to
-// LDS once per block in torch Linear layout [out][in] (which is exactly
-// the B-fragment's k-contiguous layout for D = A @ W^T).
+// Occupancy design (PMC-driven): the first version staged W1/W2/W3 in LDS
+// (81 KB) and ran at 1 block/CU with 65% of wave cycles parked in
+// SQ_WAIT_ANY. B-fragments are 16-byte k-contiguous reads, so the weights
+// (<=40 KB, L2-resident, shared by every block) are now read DIRECTLY from
+// global memory from row-padded copies prepared in the launcher; LDS holds
+// only the per-tile input/activation buffers (~43 KB -> 3 blocks/CU).
 
 #include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
@@ -28,14 +25,14 @@
 
 namespace {
 
-constexpr int H = 64;          // hidden_nf (kernel is specialized for 64)
-constexpr int EA = 2;          // edge_attr_nf
+constexpr int H = 64;
+constexpr int EA = 2;
 constexpr int K_IN = 2 * H + 1 + EA;  // 131
-constexpr int K_PAD = 160;     // 5 MFMA k-steps of 32
-constexpr int K_STRIDE = 168;  // LDS row stride (bank-conflict-free b128)
-constexpr int H_STRIDE = 72;   // LDS stride for H-wide tiles
-constexpr int TILE = 64;       // edges per block
-constexpr int THREADS = 256;   // 4 waves
+constexpr int K_PAD = 160;            // 5 MFMA k-steps of 32
+constexpr int K_STRIDE = 168;         // LDS row stride (bank-free b128)
+constexpr int H_STRIDE = 72;
+constexpr int TILE = 64;
+constexpr int THREADS = 256;
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
@@ -44,23 +41,14 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 __device__ __forceinline__ float silu(float x) {
   return x / (1.f + __expf(-x));
 }
-__device__ __forceinline__ float dsilu(float x) {  // d/dx SiLU(x)
-  float s = 1.f / (1.f + __expf(-x));
-  return s * (1.f + x * (1.f - s));
-}
 
-// ---- LDS layout (single dynamic allocation, 16B-aligned carves) ----------
 struct SmemLayout {
-  // offsets in bytes
-  int in_tile;   // [TILE][K_STRIDE] bf16
-  int w1;        // [H][K_STRIDE]   bf16 (torch [out][in], padded)
-  int w2;        // [H][H_STRIDE]   bf16
-  int w3;        // [H][H_STRIDE]   bf16
-  int t1;        // [TILE][H_STRIDE] bf16
-  int msg;       // [TILE][H_STRIDE] bf16
-  int diff;      // [TILE][4] float (xyz + radial_raw)
-  int pvec;      // [TILE] float
-  int bias;      // [3*H + H] float (b1,b2,b3,w3v)
+  int in_tile;  // [TILE][K_STRIDE] bf16
+  int t1;       // [TILE][H_STRIDE] bf16
+  int msg;      // [TILE][H_STRIDE] bf16
+  int diff;     // [TILE][4] f32
+  int pvec;     // [TILE] f32
+  int bias;     // [4*H] f32
   int total;
 };
 
@@ -68,9 +56,6 @@ __host__ __device__ constexpr SmemLayout smem_layout() {
   SmemLayout L{};
   int o = 0;
   L.in_tile = o; o += TILE * K_STRIDE * 2;
-  L.w1 = o; o += H * K_STRIDE * 2;
-  L.w2 = o; o += H * H_STRIDE * 2;
-  L.w3 = o; o += H * H_STRIDE * 2;
   L.t1 = o; o += TILE * H_STRIDE * 2;
   L.msg = o; o += TILE * H_STRIDE * 2;
   L.diff = o; o += TILE * 4 * 4;
@@ -80,73 +65,53 @@ __host__ __device__ constexpr SmemLayout smem_layout() {
   return L;
 }
 
-__device__ __forceinline__ bf16x8 lds_read8(const char* smem, int byte_off) {
-  return *reinterpret_cast<const bf16x8*>(smem + byte_off);
+__device__ __forceinline__ bf16x8 lds8(const char* smem, int off) {
+  return *reinterpret_cast<const bf16x8*>(smem + off);
+}
+__device__ __forceinline__ bf16x8 g8(const bf16* p) {
+  return *reinterpret_cast<const bf16x8*>(p);
+}
+// prevent LICM from hoisting all phases' weight fragments into registers
+__device__ __forceinline__ const bf16* opaque(const bf16* p) {
+  asm volatile("" : "+v"(p));
+  return p;
 }
 
-// One 16(edge)x64(out) GEMM over LDS tiles: A [TILE][a_stride] bf16 rows
-// a_row0.., B (weights) [64][b_stride] with k contiguous, ksteps of 32.
-// acc[nt] — 4 accumulators of f32x4 (C layout col=l&15(+16nt),
-// row=(l>>4)*4+r).
-__device__ __forceinline__ void mfma_16x64(const char* smem, int a_off,
-                                           int a_stride, int b_off,
-                                           int b_stride, int ksteps,
-                                           int lane, f32x4 acc[4]) {
-  int arow = lane & 15;
-  int kbase = (lane >> 4) * 8;
-  for (int kk = 0; kk < ksteps; ++kk) {
-    int k = kk * 32 + kbase;
-    bf16x8 a = lds_read8(smem, a_off + arow * a_stride + k * 2);
+// 16(edges) x 64 GEMM: A from LDS, B (weights, k-contig rows [64][kb])
+// from GLOBAL (L2-resident).
+template <int KSTEPS>
+__device__ __forceinline__ void mm_a_lds(const char* smem, int a_off,
+                                         int a_stride,
+                                         const bf16* __restrict__ w, int wk,
+                                         int lane, f32x4 (&acc)[4]) {
+#pragma unroll
+  for (int kk = 0; kk < KSTEPS; ++kk) {
+    int k = kk * 32 + (lane >> 4) * 8;
+    bf16x8 a = lds8(smem, a_off + (lane & 15) * a_stride + k * 2);
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 b = lds_read8(smem,
-                           b_off + (nt * 16 + (lane & 15)) * b_stride + k * 2);
+      bf16x8 b = g8(w + (nt * 16 + (lane & 15)) * wk + k);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
 }
 
-// Stage a [H][in_w] weight matrix into LDS rows of byte-stride `stride`
-// (zero-padding columns in_w..pad_w).
-__device__ __forceinline__ void stage_weight(const bf16* __restrict__ w,
-                                             char* smem, int off, int in_w,
-                                             int pad_w, int stride, int tid) {
-  for (int idx = tid; idx < H * pad_w / 8; idx += THREADS) {
-    int r = idx / (pad_w / 8);
-    int c8 = (idx % (pad_w / 8)) * 8;
-    bf16x8 v = {};
-#pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      int c = c8 + u;
-      v[u] = (c < in_w) ? ((const __bf16*)w)[r * in_w + c] : (__bf16)0.f;
-    }
-    *reinterpret_cast<bf16x8*>(smem + off + r * stride + c8 * 2) = v;
-  }
-}
-
-__global__ __launch_bounds__(THREADS) void fused_edge_fwd(
-    const bf16* __restrict__ h,        // [N,64]
-    const float* __restrict__ coord,   // [N,3]
-    const float* __restrict__ eattr,   // [M,EA]
-    const long* __restrict__ row,      // [M]
-    const long* __restrict__ col,      // [M]
-    const bf16* __restrict__ w1, const float* __restrict__ b1,
-    const bf16* __restrict__ w2, const float* __restrict__ b2,
-    const bf16* __restrict__ w3, const float* __restrict__ b3,
-    const float* __restrict__ w3v,     // [64] head vector
-    bf16* __restrict__ msg_out,        // [M,64]
-    float* __restrict__ trans_out,     // [M,3]
-    long m, int normalize, float eps) {
+__global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
+    const bf16* __restrict__ h, const float* __restrict__ coord,
+    const float* __restrict__ eattr, const long* __restrict__ row,
+    const long* __restrict__ col,
+    const bf16* __restrict__ w1p,  // [64][K_PAD] padded
+    const float* __restrict__ b1, const bf16* __restrict__ w2,
+    const float* __restrict__ b2, const bf16* __restrict__ w3,
+    const float* __restrict__ b3, const float* __restrict__ w3v,
+    bf16* __restrict__ msg_out, float* __restrict__ trans_out, long m,
+    int normalize, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr SmemLayout L = smem_layout();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  // stage weights + biases once per block
-  stage_weight(w1, smem, L.w1, K_IN, K_PAD, K_STRIDE * 2, tid);
-  stage_weight(w2, smem, L.w2, H, H, H_STRIDE * 2, tid);
-  stage_weight(w3, smem, L.w3, H, H, H_STRIDE * 2, tid);
   float* biases = reinterpret_cast<float*>(smem + L.bias);
   for (int i = tid; i < H; i += THREADS) {
     biases[i] = b1[i];
@@ -158,31 +123,25 @@ __global__ __launch_bounds__(THREADS) void fused_edge_fwd(
   for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
-    __syncthreads();  // protect LDS reuse across tiles
+    __syncthreads();
 
-    // ---- gather stage: build in_tile [64][K_STRIDE] ----
-    // 4 threads per edge: t covers h_i/h_j in 16B pieces.
+    // gather stage: in_tile [64][K_STRIDE]
     for (int idx = tid; idx < TILE * 16; idx += THREADS) {
-      int e = idx / 16;       // edge in tile
-      int piece = idx % 16;   // 16 x 16B pieces = 2 rows of 64 bf16
+      int e = idx / 16, piece = idx % 16;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
+      int c8 = (piece & 7) * 8;
+      bf16x8 v = {};
       if (e < nedge) {
         long ge = e0 + e;
-        long src_node = piece < 8 ? row[ge] : col[ge];
-        int c8 = (piece & 7) * 8;
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(h + src_node * H + c8);
-        *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
-      } else {
-        int c8 = (piece & 7) * 8;
-        bf16x8 z = {};
-        *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = z;
+        long src = piece < 8 ? row[ge] : col[ge];
+        v = g8(h + src * H + c8);
       }
+      *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
     }
-    // radial / coord_diff / edge_attr + zero K padding (1 thread per edge)
     for (int e = tid; e < TILE; e += THREADS) {
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
-      float dx = 0.f, dy = 0.f, dz = 0.f, r2 = 0.f, a0 = 0.f, a1 = 0.f;
+      float dx = 0, dy = 0, dz = 0, r2 = 0, a0 = 0, a1 = 0;
       if (e < nedge) {
         long ge = e0 + e;
         long i = row[ge], j = col[ge];
@@ -207,11 +166,10 @@ __global__ __launch_bounds__(THREADS) void fused_edge_fwd(
     }
     __syncthreads();
 
-    // ---- GEMM1: t1 = SiLU(in @ W1^T + b1) ----
-    {
+    {  // GEMM1 -> t1
       f32x4 acc[4] = {};
-      mfma_16x64(smem, L.in_tile + wave * 16 * K_STRIDE * 2, K_STRIDE * 2,
-                 L.w1, K_STRIDE * 2, K_PAD / 32, lane, acc);
+      mm_a_lds<K_PAD / 32>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
+                           K_STRIDE * 2, opaque(w1p), K_PAD, lane, acc);
       __bf16* t1 = reinterpret_cast<__bf16*>(smem + L.t1);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -224,12 +182,10 @@ __global__ __launch_bounds__(THREADS) void fused_edge_fwd(
       }
     }
     __syncthreads();
-
-    // ---- GEMM2: msg = SiLU(t1 @ W2^T + b2) ----
-    {
+    {  // GEMM2 -> msg
       f32x4 acc[4] = {};
-      mfma_16x64(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2, L.w2,
-                 H_STRIDE * 2, H / 32, lane, acc);
+      mm_a_lds<2>(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                  opaque(w2), H, lane, acc);
       __bf16* mg = reinterpret_cast<__bf16*>(smem + L.msg);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -242,49 +198,38 @@ __global__ __launch_bounds__(THREADS) void fused_edge_fwd(
       }
     }
     __syncthreads();
-
-    // ---- GEMM3 + head: p = SiLU(msg @ W3^T + b3) . w3v ----
-    {
+    {  // GEMM3 + head -> p
       f32x4 acc[4] = {};
-      mfma_16x64(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2, L.w3,
-                 H_STRIDE * 2, H / 32, lane, acc);
-      float part[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) part[r] = 0.f;
+      mm_a_lds<2>(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                  opaque(w3), H, lane, acc);
+      float part[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
         float wv = biases[3 * H + c];
         float bb = biases[2 * H + c];
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          part[r] += silu(acc[nt][r] + bb) * wv;
+        for (int r = 0; r < 4; ++r) part[r] += silu(acc[nt][r] + bb) * wv;
       }
-      // reduce the 16 lanes of each quarter-wave (same rows)
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) {
+      for (int off = 1; off < 16; off <<= 1)
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          part[r] += __shfl_xor(part[r], off, 64);
-      }
+        for (int r = 0; r < 4; ++r) part[r] += __shfl_xor(part[r], off, 64);
       if ((lane & 15) == 0) {
         float* pv = reinterpret_cast<float*>(smem + L.pvec);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int e = wave * 16 + (lane >> 4) * 4 + r;
-          pv[e] = part[r];
-        }
+        for (int r = 0; r < 4; ++r)
+          pv[wave * 16 + (lane >> 4) * 4 + r] = part[r];
       }
     }
     __syncthreads();
 
-    // ---- write msg (coalesced from LDS) + trans ----
     for (int idx = tid; idx < TILE * 8; idx += THREADS) {
       int e = idx / 8;
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
-      bf16x8 v = lds_read8(smem, L.msg + (e * H_STRIDE + c8) * 2);
-      *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) = v;
+      *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
+          lds8(smem, L.msg + (e * H_STRIDE + c8) * 2);
     }
     for (int e = tid; e < nedge; e += THREADS) {
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
@@ -297,24 +242,22 @@ __global__ __launch_bounds__(THREADS) void fused_edge_fwd(
 }
 
 // ---- MFMA layout probe (test harness): D = A[16x32] @ B[32x16] ----------
-__global__ void mfma_probe_kernel(const bf16* __restrict__ a,   // [16][32]
-                                  const bf16* __restrict__ bt,  // [16][32] B^T
-                                  float* __restrict__ d) {      // [16][16]
+__global__ void mfma_probe_kernel(const bf16* __restrict__ a,
+                                  const bf16* __restrict__ bt,
+                                  float* __restrict__ d) {
   int lane = threadIdx.x & 63;
   bf16x8 av, bv;
 #pragma unroll
   for (int u = 0; u < 8; ++u) {
     int r = lane & 15, k = (lane >> 4) * 8 + u;
     av[u] = ((const __bf16*)a)[r * 32 + k];
-    bv[u] = ((const __bf16*)bt)[r * 32 + k];  // bt[col][k] = B[k][col]
+    bv[u] = ((const __bf16*)bt)[r * 32 + k];
   }
   f32x4 acc = {};
   acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc, 0, 0, 0);
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int rr = (lane >> 4) * 4 + r, cc = lane & 15;
-    d[rr * 16 + cc] = acc[r];
-  }
+  for (int r = 0; r < 4; ++r)
+    d[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
 }
 
 }  // namespace
@@ -338,8 +281,9 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
   auto stream = at::hip::getCurrentHIPStream();
   constexpr SmemLayout L = smem_layout();
   long tiles = (m + TILE - 1) / TILE;
-  int blocks = (int)std::min<long>(tiles, 8192);
-  auto w1c = w1.contiguous();
+  int blocks = (int)std::min<long>(tiles, 16384);
+  // row-pad W1 to [64][K_PAD] so 16-B B-fragment reads are aligned
+  auto w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - K_IN});
   auto w2c = w2.contiguous();
   auto w3c = w3.contiguous();
   auto b1c = b1.contiguous().to(torch::kFloat);
@@ -350,7 +294,7 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
       reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
       col.contiguous().data_ptr<long>(),
-      reinterpret_cast<const bf16*>(w1c.data_ptr()), b1c.data_ptr<float>(),
+      reinterpret_cast<const bf16*>(w1p.data_ptr()), b1c.data_ptr<float>(),
       reinterpret_cast<const bf16*>(w2c.data_ptr()), b2c.data_ptr<float>(),
       reinterpret_cast<const bf16*>(w3c.data_ptr()), b3c.data_ptr<float>(),
       w3vc.data_ptr<float>(),
@@ -362,8 +306,7 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt) {
   auto ac = a.contiguous();
   auto btc = bt.contiguous();
-  auto d = torch::empty({16, 16},
-                        a.options().dtype(torch::kFloat));
+  auto d = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   mfma_probe_kernel<<<1, 64, 0, stream>>>(
       reinterpret_cast<const bf16*>(ac.data_ptr()),

@@ -1,27 +1,23 @@
 // Fused backward of the FastEGNN edge block (gfx950).
 //
-// Recomputes the forward chain tile-by-tile in LDS (checkpoint style) and
-// produces every per-edge gradient in ONE kernel:
-//   inputs : h, coord, eattr, row, col, weights, and the node-level
-//            cotangents already divided by degree (dmsg_n [N,64] bf16,
-//            dtrans_n [N,3] f32 — d(agg)/deg so the segment-mean backward
-//            is a plain gather).
-//   outputs: per-edge activations for the python-side wgrad GEMMs
-//            (ein [M,144] bf16, t1/msg [M,64] bf16, dz1/dz2/dz3 [M,64]
-//            bf16), per-edge input grads (dh_row/dh_col [M,64] bf16,
-//            dcd [M,3] f32 = full gradient w.r.t. the RAW coordinate
-//            difference, radial + normalize terms folded in), and the
-//            head-vector grad dw3v (per-block LDS partial + 64 atomics).
-// The caller then runs: dW_k = dz_k^T @ {ein,t1,msg} (3 library GEMMs),
-// db_k = dz_k.sum(0), and CSR segment sums for dh/dcoord — no index_add
-// scatters, no [M,.] autograd graph.
+// Recomputes the forward chain tile-by-tile (checkpoint style) and emits
+// every per-edge gradient in ONE kernel. See fused_edge.hip for the chain;
+// the python side finishes with three split-K wgrad GEMMs, bias column
+// sums and CSR segment sums (ops/__init__.py _FusedEdgeBlockFn.backward).
 //
-// Chain (see fused_edge.hip for the forward):
+// Occupancy design (PMC-driven, same as the forward): weights are read
+// from GLOBAL padded/transposed copies (L2-resident) instead of LDS, and
+// activation/derivative tiles are consolidated into PRE-ACTIVATION tiles
+// (silu / silu' computed on the fly) that are overwritten in place by the
+// dz chain: LDS drops from 158 KB (1 block/CU, 65% SQ_WAIT_ANY) to
+// ~53 KB, and the dein product is split into 3 register passes to stay
+// under the 2-waves/SIMD VGPR budget.
+//
 //   dp   = dtrans . cdu            dcdu = p * dtrans
 //   dz3  = (dp w3v) silu'(z3)      dw3v += sum dp s3
 //   dz2  = (dmsg_n[row] + dz3 W3) silu'(z2)
 //   dz1  = (dz2 W2) silu'(z1)
-//   dein = dz1 W1 -> dh_i, dh_j, dr2 (dea dropped: edge_attr is data)
+//   dein = dz1 W1 -> dh_i, dh_j, dr2
 //   dcd  = (normalize ? dcdu/(|d|+eps) : dcdu) + 2 d dr2
 
 #include <ATen/hip/HIPContext.h>
@@ -33,10 +29,10 @@ namespace {
 
 constexpr int H = 64;
 constexpr int EA = 2;
-constexpr int K_IN = 2 * H + 1 + EA;   // 131
+constexpr int K_IN = 2 * H + 1 + EA;  // 131
 constexpr int K_PAD = 160;
 constexpr int K_STRIDE = 168;
-constexpr int K_OUT = 144;             // padded ein rows written to global
+constexpr int K_OUT = 144;  // ein cols written to global (for dW1 GEMM)
 constexpr int H_STRIDE = 72;
 constexpr int TILE = 64;
 constexpr int THREADS = 256;
@@ -54,23 +50,14 @@ __device__ __forceinline__ float dsilu_(float x) {
 }
 
 struct Smem {
-  int in_tile;  // [TILE][K_STRIDE] bf16  ein (reused for dz1 staging)
-  int w1;       // [H][K_STRIDE]  bf16    W1 [out][in]
-  int w1t;      // [K_PAD][H_STRIDE] bf16 W1^T [in][out]
-  int w2;       // [H][H_STRIDE]
-  int w2t;      // [H][H_STRIDE]
-  int w3;       // [H][H_STRIDE]
-  int w3t;      // [H][H_STRIDE]
-  int t1;       // [TILE][H_STRIDE] bf16  (reused for dz2)
-  int msg;      // [TILE][H_STRIDE] bf16  (reused for dz3)
-  int s3;       // [TILE][H_STRIDE] bf16
-  int ds1;      // [TILE][H_STRIDE] bf16  silu'(z1)
-  int ds2;      // [TILE][H_STRIDE] bf16
-  int ds3;      // [TILE][H_STRIDE] bf16
-  int diff;     // [TILE][4] f32 (raw dx,dy,dz,r2)
-  int scal;     // [TILE][4] f32 (p, dp, unused, unused)
-  int bias;     // [4*H] f32 (b1,b2,b3,w3v)
-  int wpart;    // [H] f32 dw3v block partial
+  int in_tile;  // [TILE][K_STRIDE] bf16  (ein, later dein)
+  int z1;       // [TILE][H_STRIDE] bf16  (pre-act; overwritten by dz1)
+  int z2;       // (overwritten by dz2)
+  int z3;       // (overwritten by dz3)
+  int diff;     // [TILE][4] f32
+  int scal;     // [TILE][4] f32 (p, dp/dr2, -, -)
+  int bias;     // [4*H] f32
+  int wpart;    // [H] f32
   int total;
 };
 
@@ -78,18 +65,9 @@ __host__ __device__ constexpr Smem smem_layout() {
   Smem L{};
   int o = 0;
   L.in_tile = o; o += TILE * K_STRIDE * 2;
-  L.w1 = o; o += H * K_STRIDE * 2;
-  L.w1t = o; o += K_PAD * H_STRIDE * 2;
-  L.w2 = o; o += H * H_STRIDE * 2;
-  L.w2t = o; o += H * H_STRIDE * 2;
-  L.w3 = o; o += H * H_STRIDE * 2;
-  L.w3t = o; o += H * H_STRIDE * 2;
-  L.t1 = o; o += TILE * H_STRIDE * 2;
-  L.msg = o; o += TILE * H_STRIDE * 2;
-  L.s3 = o; o += TILE * H_STRIDE * 2;
-  L.ds1 = o; o += TILE * H_STRIDE * 2;
-  L.ds2 = o; o += TILE * H_STRIDE * 2;
-  L.ds3 = o; o += TILE * H_STRIDE * 2;
+  L.z1 = o; o += TILE * H_STRIDE * 2;
+  L.z2 = o; o += TILE * H_STRIDE * 2;
+  L.z3 = o; o += TILE * H_STRIDE * 2;
   L.diff = o; o += TILE * 4 * 4;
   L.scal = o; o += TILE * 4 * 4;
   L.bias = o; o += 4 * H * 4;
@@ -98,101 +76,72 @@ __host__ __device__ constexpr Smem smem_layout() {
   return L;
 }
 
-__device__ __forceinline__ bf16x8 rd8(const char* smem, int off) {
+__device__ __forceinline__ bf16x8 lds8(const char* smem, int off) {
   return *reinterpret_cast<const bf16x8*>(smem + off);
 }
+__device__ __forceinline__ bf16x8 g8(const bf16* p) {
+  return *reinterpret_cast<const bf16x8*>(p);
+}
+// Launder a pointer so the compiler cannot hoist its loads across phases:
+// without this, LICM pre-loads EVERY phase's weight fragments into
+// registers across the tile loop (measured 256 VGPR + 170 AGPR -> 1
+// wave/SIMD).
+__device__ __forceinline__ const bf16* opaque(const bf16* p) {
+  asm volatile("" : "+v"(p));
+  return p;
+}
+// read 8 bf16 pre-activations from LDS and apply silu -> bf16x8
+__device__ __forceinline__ bf16x8 lds8_silu(const char* smem, int off) {
+  bf16x8 z = lds8(smem, off);
+  bf16x8 r;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) r[u] = (__bf16)silu_((float)z[u]);
+  return r;
+}
 
-// A [TILE rows from a_off][a_stride], B [n16*16+col][b_stride] k-contig.
-// Compile-time KSTEPS/NT: runtime-indexed ext_vector arrays would spill to
-// scratch (guide 5.4 rule 20).
-template <int KSTEPS, int NT>
-__device__ __forceinline__ void mm_16xN(const char* smem, int a_off,
-                                        int a_stride, int b_off, int b_stride,
-                                        int lane, f32x4 (&acc)[NT]) {
+// A from LDS (optionally through silu), B from global [64][wk] k-contig.
+template <int KSTEPS, bool SILU_A>
+__device__ __forceinline__ void mm_g(const char* smem, int a_off,
+                                     int a_stride,
+                                     const bf16* __restrict__ w, int wk,
+                                     int lane, f32x4 (&acc)[4]) {
 #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
-    bf16x8 a = rd8(smem, a_off + (lane & 15) * a_stride + k * 2);
+    bf16x8 a = SILU_A ? lds8_silu(smem, a_off + (lane & 15) * a_stride + k * 2)
+                      : lds8(smem, a_off + (lane & 15) * a_stride + k * 2);
 #pragma unroll
-    for (int nt = 0; nt < NT; ++nt) {
-      bf16x8 b = rd8(smem, b_off + (nt * 16 + (lane & 15)) * b_stride + k * 2);
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8 b = g8(w + (nt * 16 + (lane & 15)) * wk + k);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
 }
 
-__device__ __forceinline__ void stage_w(const bf16* __restrict__ w, char* smem,
-                                        int off, int rows, int in_w, int pad_w,
-                                        int stride, int tid, bool transpose,
-                                        int t_rows) {
-  // transpose=false: LDS[r][c] = w[r*in_w + c] for r<rows
-  // transpose=true : LDS[r][c] = w[c*in_w + r] (stage W^T; r<t_rows, c<rows)
-  int nrow = transpose ? t_rows : rows;
-  for (int idx = tid; idx < nrow * pad_w / 8; idx += THREADS) {
-    int r = idx / (pad_w / 8);
-    int c8 = (idx % (pad_w / 8)) * 8;
-    bf16x8 v = {};
-#pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      int c = c8 + u;
-      if (!transpose) {
-        v[u] = (c < in_w) ? ((const __bf16*)w)[r * in_w + c] : (__bf16)0.f;
-      } else {
-        v[u] = (c < rows && r < in_w) ? ((const __bf16*)w)[c * in_w + r]
-                                      : (__bf16)0.f;
-      }
-    }
-    *reinterpret_cast<bf16x8*>(smem + off + r * stride + c8 * 2) = v;
-  }
-}
-
-// write a wave's 16x64 C tile (4 f32x4 accs) into an LDS bf16 tile,
-// optionally applying f(x) per element; C layout col=l&15+16nt,
-// row=(l>>4)*4+r.
-#define WRITE_TILE(dst_off, stride, expr)                                   \
-  do {                                                                      \
-    __bf16* _d = reinterpret_cast<__bf16*>(smem + (dst_off));               \
-    _Pragma("unroll") for (int nt = 0; nt < 4; ++nt) {                      \
-      int c = nt * 16 + (lane & 15);                                        \
-      _Pragma("unroll") for (int r = 0; r < 4; ++r) {                       \
-        int e = wave * 16 + (lane >> 4) * 4 + r;                            \
-        float x = acc[nt][r];                                               \
-        _d[e * (stride) + c] = (__bf16)(expr);                              \
-      }                                                                     \
-    }                                                                       \
-  } while (0)
-
 __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
     const bf16* __restrict__ h, const float* __restrict__ coord,
     const float* __restrict__ eattr, const long* __restrict__ row,
     const long* __restrict__ col,
-    const bf16* __restrict__ dmsg_n,    // [N,64] dagg_msg/deg
-    const float* __restrict__ dtrans_n,  // [N,3] dagg_trans/deg
-    const bf16* __restrict__ w1, const float* __restrict__ b1,
-    const bf16* __restrict__ w2, const float* __restrict__ b2,
-    const bf16* __restrict__ w3, const float* __restrict__ b3,
-    const float* __restrict__ w3v,
-    bf16* __restrict__ ein_out,    // [M,K_OUT]
-    bf16* __restrict__ t1_out,     // [M,64]
-    bf16* __restrict__ msg_out,    // [M,64]
-    bf16* __restrict__ dz1_out, bf16* __restrict__ dz2_out,
-    bf16* __restrict__ dz3_out,   // [M,64] each
-    bf16* __restrict__ dhr_out, bf16* __restrict__ dhc_out,  // [M,64]
-    float* __restrict__ dcd_out,  // [M,3]
-    float* __restrict__ dw3v_out,  // [64]
-    long m, int normalize, float eps) {
+    const bf16* __restrict__ dmsg_n,     // [N,64]
+    const float* __restrict__ dtrans_n,  // [N,3]
+    const bf16* __restrict__ w1p,   // [64][K_PAD]
+    const bf16* __restrict__ w1tp,  // [K_OUT][64] (W1^T rows padded)
+    const bf16* __restrict__ w2, const bf16* __restrict__ w2t,
+    const bf16* __restrict__ w3, const bf16* __restrict__ w3t,
+    const float* __restrict__ b1, const float* __restrict__ b2,
+    const float* __restrict__ b3, const float* __restrict__ w3v,
+    bf16* __restrict__ ein_out, bf16* __restrict__ t1_out,
+    bf16* __restrict__ msg_out, bf16* __restrict__ dz1_out,
+    bf16* __restrict__ dz2_out, bf16* __restrict__ dz3_out,
+    bf16* __restrict__ dhr_out, bf16* __restrict__ dhc_out,
+    float* __restrict__ dcd_out, float* __restrict__ dw3v_out, long m,
+    int normalize, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr Smem L = smem_layout();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  stage_w(w1, smem, L.w1, H, K_IN, K_PAD, K_STRIDE * 2, tid, false, 0);
-  stage_w(w1, smem, L.w1t, H, K_IN, H, H_STRIDE * 2, tid, true, K_PAD);
-  stage_w(w2, smem, L.w2, H, H, H, H_STRIDE * 2, tid, false, 0);
-  stage_w(w2, smem, L.w2t, H, H, H, H_STRIDE * 2, tid, true, H);
-  stage_w(w3, smem, L.w3, H, H, H, H_STRIDE * 2, tid, false, 0);
-  stage_w(w3, smem, L.w3t, H, H, H, H_STRIDE * 2, tid, true, H);
   float* biases = reinterpret_cast<float*>(smem + L.bias);
   float* wpart = reinterpret_cast<float*>(smem + L.wpart);
   for (int i = tid; i < H; i += THREADS) {
@@ -208,7 +157,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
     __syncthreads();
 
-    // ---- stage ein (identical to forward gather) ----
+    // ---- stage ein ----
     for (int idx = tid; idx < TILE * 16; idx += THREADS) {
       int e = idx / 16, piece = idx % 16;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
@@ -217,7 +166,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       if (e < nedge) {
         long ge = e0 + e;
         long src = piece < 8 ? row[ge] : col[ge];
-        v = *reinterpret_cast<const bf16x8*>(h + src * H + c8);
+        v = g8(h + src * H + c8);
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
     }
@@ -244,73 +193,79 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       for (int k = K_IN; k < K_PAD; ++k) brow[k] = (__bf16)0.f;
     }
     __syncthreads();
-
-    // ein -> global (K_OUT cols)
+    // ein -> global
     for (int idx = tid; idx < TILE * (K_OUT / 8); idx += THREADS) {
       int e = idx / (K_OUT / 8);
       if (e >= nedge) continue;
       int c8 = (idx % (K_OUT / 8)) * 8;
       *reinterpret_cast<bf16x8*>(ein_out + (e0 + e) * K_OUT + c8) =
-          rd8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
+          lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
     }
 
-    // ---- recompute forward: t1, msg, s3 (+ silu' tiles) ----
+    // ---- recompute: z1, z2, z3 (pre-activations) ----
     {
       f32x4 acc[4] = {};
-      mm_16xN<K_PAD / 32, 4>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
-                             K_STRIDE * 2, L.w1, K_STRIDE * 2, lane, acc);
-      __bf16* d1 = reinterpret_cast<__bf16*>(smem + L.ds1);
+      mm_g<K_PAD / 32, false>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
+                              K_STRIDE * 2, opaque(w1p), K_PAD, lane, acc);
+      __bf16* z1 = reinterpret_cast<__bf16*>(smem + L.z1);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
-          float z = acc[nt][r] + biases[c];
-          reinterpret_cast<__bf16*>(smem + L.t1)[e * H_STRIDE + c] =
-              (__bf16)silu_(z);
-          d1[e * H_STRIDE + c] = (__bf16)dsilu_(z);
+          z1[e * H_STRIDE + c] = (__bf16)(acc[nt][r] + biases[c]);
         }
       }
     }
     __syncthreads();
+    // t1 = silu(z1) -> global (coalesced)
+    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+      int e = idx / 8;
+      if (e >= nedge) continue;
+      int c8 = (idx % 8) * 8;
+      *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
+          lds8_silu(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+    }
     {
       f32x4 acc[4] = {};
-      mm_16xN<2, 4>(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                    L.w2, H_STRIDE * 2, lane, acc);
-      __bf16* d2 = reinterpret_cast<__bf16*>(smem + L.ds2);
+      mm_g<2, true>(smem, L.z1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                    opaque(w2), H, lane, acc);
+      __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
-          float z = acc[nt][r] + biases[H + c];
-          reinterpret_cast<__bf16*>(smem + L.msg)[e * H_STRIDE + c] =
-              (__bf16)silu_(z);
-          d2[e * H_STRIDE + c] = (__bf16)dsilu_(z);
+          z2[e * H_STRIDE + c] = (__bf16)(acc[nt][r] + biases[H + c]);
         }
       }
     }
     __syncthreads();
+    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+      int e = idx / 8;
+      if (e >= nedge) continue;
+      int c8 = (idx % 8) * 8;
+      *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
+          lds8_silu(smem, L.z2 + (e * H_STRIDE + c8) * 2);
+    }
     {
       f32x4 acc[4] = {};
-      mm_16xN<2, 4>(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                    L.w3, H_STRIDE * 2, lane, acc);
-      __bf16* d3 = reinterpret_cast<__bf16*>(smem + L.ds3);
+      mm_g<2, true>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                    opaque(w3), H, lane, acc);
+      __bf16* z3 = reinterpret_cast<__bf16*>(smem + L.z3);
       float part[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
         float wv = biases[3 * H + c];
+        float bb = biases[2 * H + c];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          int e = wave * 16 + (lane >> 4) * 4 + r;
-          float z = acc[nt][r] + biases[2 * H + c];
-          float s = silu_(z);
-          reinterpret_cast<__bf16*>(smem + L.s3)[e * H_STRIDE + c] = (__bf16)s;
-          d3[e * H_STRIDE + c] = (__bf16)dsilu_(z);
-          part[r] += s * wv;
+          float z = acc[nt][r] + bb;
+          z3[((wave * 16 + (lane >> 4) * 4 + r)) * H_STRIDE + c] = (__bf16)z;
+          part[r] += silu_(z) * wv;
         }
       }
 #pragma unroll
@@ -326,83 +281,53 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
     }
     __syncthreads();
 
-    // write t1/msg global (for wgrad GEMMs)
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
-          rd8(smem, L.t1 + (e * H_STRIDE + c8) * 2);
-      *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
-          rd8(smem, L.msg + (e * H_STRIDE + c8) * 2);
-    }
-
-    // ---- head backward: dp, dcdu; dcd written; dw3v partial ----
+    // ---- head backward: dp; dw3v partial; dz3 overwrites z3 ----
     for (int e = tid; e < TILE; e += THREADS) {
       float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
-      float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
+      const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float dp = 0.f;
       if (e < nedge) {
         long ge = e0 + e;
         long i = row[ge];
-        float tx = dtrans_n[i * 3], ty = dtrans_n[i * 3 + 1],
-              tz = dtrans_n[i * 3 + 2];
-        float dx = dptr[0], dy = dptr[1], dz = dptr[2], r2 = dptr[3];
-        float inv = normalize ? 1.f / (sqrtf(r2) + eps) : 1.f;
-        float cx = dx * inv, cy = dy * inv, cz = dz * inv;
-        float p = sc[0];
-        dp = tx * cx + ty * cy + tz * cz;
-        // dcdu = p * dtrans; fold normalize + radial term later (needs dr2)
-        (void)p;
-        sc[1] = dp;
-      } else {
-        sc[1] = 0.f;
+        float inv = normalize ? 1.f / (sqrtf(dptr[3]) + eps) : 1.f;
+        dp = (dtrans_n[i * 3] * dptr[0] + dtrans_n[i * 3 + 1] * dptr[1] +
+              dtrans_n[i * 3 + 2] * dptr[2]) * inv;
       }
+      sc[1] = dp;
     }
     __syncthreads();
-
-    // dz3 = (dp (x) w3v) * silu'(z3) into the (freed) msg tile;
-    // dw3v_part[c] += sum_e dp[e] * s3[e][c]
     {
-      // per-column dw3v partial: thread covers (e strip, c)
-      // layout: 256 threads = 4 waves; each thread handles c = tid%64 over
-      // 16 edges
+      // thread covers column c over 16 edges: dw3v partial + dz3 in place
       int c = tid & 63;
       int estart = (tid >> 6) * 16;
       float acc_w = 0.f;
-      const __bf16* s3p = reinterpret_cast<const __bf16*>(smem + L.s3);
-      const float* sc = reinterpret_cast<const float*>(smem + L.scal);
-      __bf16* dz3t = reinterpret_cast<__bf16*>(smem + L.msg);  // reuse msg
-      const __bf16* d3 = reinterpret_cast<const __bf16*>(smem + L.ds3);
       float wv = biases[3 * H + c];
+      __bf16* z3 = reinterpret_cast<__bf16*>(smem + L.z3);
+      const float* sc = reinterpret_cast<const float*>(smem + L.scal);
       for (int e = estart; e < estart + 16; ++e) {
+        float z = (float)z3[e * H_STRIDE + c];
         float dp = sc[e * 4 + 1];
-        acc_w += dp * (float)s3p[e * H_STRIDE + c];
-        dz3t[e * H_STRIDE + c] =
-            (__bf16)(dp * wv * (float)d3[e * H_STRIDE + c]);
+        acc_w += dp * silu_(z);
+        z3[e * H_STRIDE + c] = (__bf16)(dp * wv * dsilu_(z));
       }
-      // accumulate into block partial (4 contributions per column)
       atomicAdd(&wpart[c], acc_w);
     }
     __syncthreads();
-
-    // dz3 -> global
     for (int idx = tid; idx < TILE * 8; idx += THREADS) {
       int e = idx / 8;
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(dz3_out + (e0 + e) * H + c8) =
-          rd8(smem, L.msg + (e * H_STRIDE + c8) * 2);
+          lds8(smem, L.z3 + (e * H_STRIDE + c8) * 2);
     }
 
-    // ---- dz2 = (dmsg_n[row] + dz3 @ W3) * silu'(z2), into t1 tile ----
+    // ---- dz2 = (dmsg_n[row] + dz3 @ W3) silu'(z2), overwrite z2 ----
     {
       f32x4 acc[4] = {};
-      mm_16xN<2, 4>(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                    L.w3t, H_STRIDE * 2, lane, acc);
+      mm_g<2, false>(smem, L.z3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                     opaque(w3t), H, lane, acc);
       __syncthreads();
-      __bf16* dz2t = reinterpret_cast<__bf16*>(smem + L.t1);
-      const __bf16* d2 = reinterpret_cast<const __bf16*>(smem + L.ds2);
+      __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
@@ -410,12 +335,10 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
           float up = 0.f;
-          if (e < nedge) {
-            long ge = e0 + e;
-            up = (float)((const __bf16*)dmsg_n)[row[ge] * H + c];
-          }
-          dz2t[e * H_STRIDE + c] =
-              (__bf16)((acc[nt][r] + up) * (float)d2[e * H_STRIDE + c]);
+          if (e < nedge)
+            up = (float)((const __bf16*)dmsg_n)[row[e0 + e] * H + c];
+          float z = (float)z2[e * H_STRIDE + c];
+          z2[e * H_STRIDE + c] = (__bf16)((acc[nt][r] + up) * dsilu_(z));
         }
       }
     }
@@ -425,25 +348,24 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
-          rd8(smem, L.t1 + (e * H_STRIDE + c8) * 2);
+          lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
     }
 
-    // ---- dz1 = (dz2 @ W2) * silu'(z1), into in_tile rows (reuse) ----
+    // ---- dz1 = (dz2 @ W2) silu'(z1), overwrite z1 ----
     {
       f32x4 acc[4] = {};
-      mm_16xN<2, 4>(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                    L.w2t, H_STRIDE * 2, lane, acc);
+      mm_g<2, false>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                     opaque(w2t), H, lane, acc);
       __syncthreads();
-      __bf16* dz1t = reinterpret_cast<__bf16*>(smem + L.s3);  // reuse s3
-      const __bf16* d1 = reinterpret_cast<const __bf16*>(smem + L.ds1);
+      __bf16* z1 = reinterpret_cast<__bf16*>(smem + L.z1);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
-          dz1t[e * H_STRIDE + c] =
-              (__bf16)(acc[nt][r] * (float)d1[e * H_STRIDE + c]);
+          float z = (float)z1[e * H_STRIDE + c];
+          z1[e * H_STRIDE + c] = (__bf16)(acc[nt][r] * dsilu_(z));
         }
       }
     }
@@ -453,56 +375,70 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
-          rd8(smem, L.s3 + (e * H_STRIDE + c8) * 2);
+          lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
     }
 
-    // ---- dein = dz1 @ W1 (9 n-tiles over K_PAD=144 cols) ----
-    {
-      f32x4 acc[9] = {};
-      mm_16xN<2, 9>(smem, L.s3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                    L.w1t, H_STRIDE * 2, lane, acc);
-      // dh_row = dein[0:64], dh_col = dein[64:128]; dr2 = dein[128]
+    // ---- dein = dz1 @ W1 in 3 register passes of 3 n-tiles ----
+    // pass outputs land in in_tile (ein no longer needed)
 #pragma unroll
-      for (int nt = 0; nt < 9; ++nt) {
-        int c = nt * 16 + (lane & 15);
+    for (int pass = 0; pass < 3; ++pass) {
+      const bf16* w1tp_ = opaque(w1tp);
+      f32x4 acc[3] = {};
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        int k = kk * 32 + (lane >> 4) * 8;
+        bf16x8 a = lds8(smem, L.z1 + (wave * 16 + (lane & 15)) * H_STRIDE * 2
+                                  + k * 2);
+#pragma unroll
+        for (int nt = 0; nt < 3; ++nt) {
+          int gc = (pass * 3 + nt) * 16 + (lane & 15);
+          bf16x8 b = g8(w1tp_ + gc * H + k);
+          acc[nt] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+        }
+      }
+      __bf16* dein = reinterpret_cast<__bf16*>(smem + L.in_tile);
+#pragma unroll
+      for (int nt = 0; nt < 3; ++nt) {
+        int c = (pass * 3 + nt) * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
-          if (e >= nedge) continue;
-          long ge = e0 + e;
-          if (c < H) {
-            dhr_out[ge * H + c] = (bf16)__float2bfloat16(acc[nt][r]);
-          } else if (c < 2 * H) {
-            dhc_out[ge * H + (c - H)] = (bf16)__float2bfloat16(acc[nt][r]);
-          } else if (c == 2 * H) {
-            // radial grad -> store into scal slot 1? combine below in dcd
+          dein[e * K_STRIDE + c] = (__bf16)acc[nt][r];
+          if (c == 2 * H) {
             float* sc = reinterpret_cast<float*>(smem + L.scal);
-            sc[e * 4 + 1] = acc[nt][r];  // overwrite dp slot with dr2
+            sc[e * 4 + 2] = acc[nt][r];  // dr2 (fp32, before bf16 rounding)
           }
         }
       }
     }
     __syncthreads();
-
-    // ---- dcd = dcdu-term + 2 d dr2 ----
-    for (int e = tid; e < TILE; e += THREADS) {
+    // dh_row / dh_col -> global (coalesced)
+    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+      int e = idx / 8;
       if (e >= nedge) continue;
+      int c8 = (idx % 8) * 8;
+      *reinterpret_cast<bf16x8*>(dhr_out + (e0 + e) * H + c8) =
+          lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
+      *reinterpret_cast<bf16x8*>(dhc_out + (e0 + e) * H + c8) =
+          lds8(smem, L.in_tile + (e * K_STRIDE + H + c8) * 2);
+    }
+    // dcd
+    for (int e = tid; e < nedge; e += THREADS) {
       long ge = e0 + e;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       const float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
       long i = row[ge];
       float tx = dtrans_n[i * 3], ty = dtrans_n[i * 3 + 1],
             tz = dtrans_n[i * 3 + 2];
-      float p = sc[0], dr2 = sc[1];
-      float dx = dptr[0], dy = dptr[1], dz = dptr[2], r2 = dptr[3];
-      float inv = normalize ? 1.f / (sqrtf(r2) + eps) : 1.f;
-      dcd_out[ge * 3] = p * tx * inv + 2.f * dx * dr2;
-      dcd_out[ge * 3 + 1] = p * ty * inv + 2.f * dy * dr2;
-      dcd_out[ge * 3 + 2] = p * tz * inv + 2.f * dz * dr2;
+      float p = sc[0], dr2 = sc[2];
+      float inv = normalize ? 1.f / (sqrtf(dptr[3]) + eps) : 1.f;
+      dcd_out[ge * 3] = p * tx * inv + 2.f * dptr[0] * dr2;
+      dcd_out[ge * 3 + 1] = p * ty * inv + 2.f * dptr[1] * dr2;
+      dcd_out[ge * 3 + 2] = p * tz * inv + 2.f * dptr[2] * dr2;
     }
   }
   __syncthreads();
-  // flush dw3v block partial
   for (int c = tid; c < H; c += THREADS) atomicAdd(&dw3v_out[c], wpart[c]);
 }
 
@@ -527,25 +463,24 @@ std::vector<torch::Tensor> fused_edge_backward(
   auto dhc = torch::empty({m, (long)H}, bopt);
   auto dcd = torch::empty({m, 3}, fopt);
   auto dw3v = torch::zeros({(long)H}, fopt);
-  if (m == 0)
-    return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v};
+  if (m == 0) return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v};
   auto stream = at::hip::getCurrentHIPStream();
   constexpr Smem L = smem_layout();
-  static bool attr_set = false;
-  if (!attr_set) {
-    hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&fused_edge_bwd),
-        hipFuncAttributeMaxDynamicSharedMemorySize, L.total);
-    attr_set = true;
-  }
   long tiles = (m + TILE - 1) / TILE;
-  int blocks = (int)std::min<long>(tiles, 8192);
+  int blocks = (int)std::min<long>(tiles, 16384);
   auto hc = h.contiguous();
   auto cc = coord.contiguous().to(torch::kFloat);
   auto ec = eattr.contiguous().to(torch::kFloat);
   auto dmn = dmsg_n.contiguous();
   auto dtn = dtrans_n.contiguous().to(torch::kFloat);
-  auto w1c = w1.contiguous(), w2c = w2.contiguous(), w3c = w3.contiguous();
+  auto w1c = w1.contiguous();
+  auto w1p = torch::constant_pad_nd(w1c, {0, K_PAD - K_IN});
+  auto w1tp = torch::constant_pad_nd(w1c.t().contiguous(),
+                                     {0, 0, 0, K_OUT - K_IN});
+  auto w2c = w2.contiguous();
+  auto w2tc = w2c.t().contiguous();
+  auto w3c = w3.contiguous();
+  auto w3tc = w3c.t().contiguous();
   auto b1c = b1.contiguous().to(torch::kFloat);
   auto b2c = b2.contiguous().to(torch::kFloat);
   auto b3c = b3.contiguous().to(torch::kFloat);
@@ -555,10 +490,13 @@ std::vector<torch::Tensor> fused_edge_backward(
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
       col.contiguous().data_ptr<long>(),
       reinterpret_cast<const bf16*>(dmn.data_ptr()), dtn.data_ptr<float>(),
-      reinterpret_cast<const bf16*>(w1c.data_ptr()), b1c.data_ptr<float>(),
-      reinterpret_cast<const bf16*>(w2c.data_ptr()), b2c.data_ptr<float>(),
-      reinterpret_cast<const bf16*>(w3c.data_ptr()), b3c.data_ptr<float>(),
-      w3vc.data_ptr<float>(),
+      reinterpret_cast<const bf16*>(w1p.data_ptr()),
+      reinterpret_cast<const bf16*>(w1tp.data_ptr()),
+      reinterpret_cast<const bf16*>(w2c.data_ptr()),
+      reinterpret_cast<const bf16*>(w2tc.data_ptr()),
+      reinterpret_cast<const bf16*>(w3c.data_ptr()),
+      reinterpret_cast<const bf16*>(w3tc.data_ptr()), b1c.data_ptr<float>(),
+      b2c.data_ptr<float>(), b3c.data_ptr<float>(), w3vc.data_ptr<float>(),
       reinterpret_cast<bf16*>(ein.data_ptr()),
       reinterpret_cast<bf16*>(t1.data_ptr()),
       reinterpret_cast<bf16*>(msg.data_ptr()),

@@ -9,7 +9,9 @@ agg = collections.defaultdict(lambda: collections.defaultdict(float))
 cnt = collections.Counter()
 for fn in files:
     for r in csv.DictReader(open(fn)):
-        k = r["Kernel_Name"].split("(")[0].split("<")[0][:60]
+        name = r["Kernel_Name"]
+        name = name.replace("(anonymous namespace)::", "")
+        k = name.split("(")[0].split("<")[0].replace("void ", "").strip()[:60]
         agg[k][r["Counter_Name"]] += float(r["Counter_Value"])
         cnt[k] = cnt[k]
         cnt[(k, r["Counter_Name"])] += 1
