@@ -62,40 +62,57 @@ def build_rank_batches(rank, world_size, num_batches, n_nodes, radius,
     return batches
 
 
-def train_step(model, batch, optimizer, grad_bucket, step, accum,
-               mmd_sigma, mmd_samples, world_size, device, autocast_dtype):
+def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
+    """forward + loss + backward on device-resident batch tensors.
+
+    hipGraph-capturable: no host syncs, no collectives (counts_global is
+    reduced eagerly by the caller before the graphed region)."""
+
+    def step_core(data):
+        total_node_cnt = data.counts_global.sum()
+        chunks = None
+        if getattr(data, "pool_chunk_begin", None) is not None:
+            chunks = (data.pool_chunk_begin, data.pool_chunk_end,
+                      data.pool_seg_chunk_ptr)
+        ctx = (torch.autocast("cuda", dtype=autocast_dtype)
+               if autocast_dtype is not None else torch.enable_grad())
+        with ctx:
+            loc_pred, vloc = model(
+                data.x, data.pos, data.vel, data.loc_mean, data.edge_index,
+                data.batch, edge_attr=data.edge_attr, node_attr=data.attr,
+                rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
+                counts_global=data.counts_global, pool_chunks=chunks,
+                colptr=data.colptr, col_perm=data.col_perm)
+        loss = torch.nn.functional.mse_loss(loc_pred.float(), data.target)
+        weight = float(data.num_nodes) / total_node_cnt
+        loss = weight * loss
+        mse_log = loss.detach()
+        lm = mmd_loss(vloc.permute(0, 2, 1).float(), data.target, data.batch,
+                      data.ptr, data.counts, mmd_sigma, mmd_samples)
+        loss = loss + 0.01 * weight * lm
+        (loss / accum).backward()
+        return (mse_log,)
+
+    return step_core
+
+
+def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
+               world_size, device):
     data = batch.to(device)
     data.counts_global = (comm.global_counts(data.counts)
                           if world_size > 1 else data.counts)
-    total_node_cnt = data.counts_global.sum()
-    chunks = None
-    if getattr(data, "pool_chunk_begin", None) is not None:
-        chunks = (data.pool_chunk_begin, data.pool_chunk_end,
-                  data.pool_seg_chunk_ptr)
-    ctx = (torch.autocast("cuda", dtype=autocast_dtype)
-           if autocast_dtype is not None else torch.enable_grad())
-    with ctx:
-        loc_pred, vloc = model(
-            data.x, data.pos, data.vel, data.loc_mean, data.edge_index,
-            data.batch, edge_attr=data.edge_attr, node_attr=data.attr,
-            rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
-            counts_global=data.counts_global, pool_chunks=chunks,
-            colptr=data.colptr, col_perm=data.col_perm)
-    loss = torch.nn.functional.mse_loss(loc_pred.float(), data.target)
-    weight = float(data.num_nodes) / total_node_cnt
-    loss = weight * loss
-    mse_log = loss.detach()
-    lm = mmd_loss(vloc.permute(0, 2, 1).float(), data.target, data.batch,
-                  data.ptr, data.counts, mmd_sigma, mmd_samples)
-    loss = loss + 0.01 * weight * lm
-    (loss / accum).backward()
+    (mse_log,) = graphed(data)
     if (step + 1) % accum == 0:
         if grad_bucket is not None:
             grad_bucket.sync()
-        torch.nn.utils.clip_grad_norm_(model.parameters(), max_norm=0.3)
+        torch.nn.utils.clip_grad_norm_(model_params(graphed), max_norm=0.3)
         optimizer.step()
         optimizer.zero_grad(set_to_none=False)
     return mse_log
+
+
+def model_params(graphed):
+    return graphed.params
 
 
 def main():
@@ -112,6 +129,9 @@ def main():
     ap.add_argument("--virtual-channels", type=int, default=5)
     ap.add_argument("--num-batches", type=int, default=2,
                     help="distinct synthetic samples to cycle through")
+    ap.add_argument("--graphs", type=str, default="on",
+                    choices=["on", "off"],
+                    help="hipGraph-capture the fwd+loss+bwd step")
     args = ap.parse_args()
 
     rank, world_size = comm.init_distributed()
@@ -146,18 +166,25 @@ def main():
     mmd_sigma, mmd_samples = 3.0, 50
     model.train()
 
+    from distegnn_amd.runtime.graphs import GraphedStep
+
+    step_core = make_step_core(model, accum, mmd_sigma, mmd_samples,
+                               autocast_dtype)
+    graphed = GraphedStep(step_core, model.parameters(),
+                          warmup_occurrences=2,
+                          enabled=(args.graphs == "on"), verbose=True)
+
     mse = None
     for w in range(args.warmup):
-        mse = train_step(model, batches[w % len(batches)], optimizer,
-                         grad_bucket, w, accum, mmd_sigma, mmd_samples,
-                         world_size, device, autocast_dtype)
+        mse = train_step(graphed, batches[w % len(batches)], optimizer,
+                         grad_bucket, w, accum, world_size, device)
     comm.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for k in range(args.steps):
-        mse = train_step(model, batches[k % len(batches)], optimizer,
-                         grad_bucket, args.warmup + k, accum, mmd_sigma,
-                         mmd_samples, world_size, device, autocast_dtype)
+        mse = train_step(graphed, batches[k % len(batches)], optimizer,
+                         grad_bucket, args.warmup + k, accum,
+                         world_size, device)
     comm.barrier()
     torch.cuda.synchronize()
     elapsed = torch.tensor(time.perf_counter() - t0, device=device)
