@@ -129,9 +129,12 @@ def main():
     ap.add_argument("--virtual-channels", type=int, default=5)
     ap.add_argument("--num-batches", type=int, default=2,
                     help="distinct synthetic samples to cycle through")
-    ap.add_argument("--graphs", type=str, default="on",
-                    choices=["on", "off"],
-                    help="hipGraph-capture the fwd+loss+bwd step")
+    ap.add_argument("--graphs", type=str, default="auto",
+                    choices=["auto", "on", "off"],
+                    help="hipGraph-capture the fwd+loss+bwd step; auto = on "
+                         "for 1 GPU, off for multi-GPU (RCCL capture of the "
+                         "in-forward virtual-node all-reduce is untested on "
+                         "this pool; force with --graphs on)")
     args = ap.parse_args()
 
     rank, world_size = comm.init_distributed()
@@ -170,9 +173,11 @@ def main():
 
     step_core = make_step_core(model, accum, mmd_sigma, mmd_samples,
                                autocast_dtype)
+    use_graphs = (args.graphs == "on"
+                  or (args.graphs == "auto" and world_size == 1))
     graphed = GraphedStep(step_core, model.parameters(),
                           warmup_occurrences=2,
-                          enabled=(args.graphs == "on"), verbose=True)
+                          enabled=use_graphs, verbose=True)
 
     mse = None
     for w in range(args.warmup):
