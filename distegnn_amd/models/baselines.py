@@ -349,3 +349,261 @@ class RF_vel(nn.Module):
             x, _ = self._modules[f"gcl_{i}"](x, vel_norm, vel, edges,
                                              edge_attr)
         return x
+
+
+class EquivariantEdgeScalarNet(nn.Module):
+    """Edge-pair O(n)-equivariant net (reference basic.py:467-506)."""
+
+    def __init__(self, n_vector_input, hidden_dim, activation,
+                 n_scalar_input=0, norm=True, flat=False):
+        super().__init__()
+        self.input_dim = n_vector_input * n_vector_input + n_scalar_input
+        self.hidden_dim = hidden_dim
+        self.output_dim = hidden_dim
+        self.norm = norm
+        self.in_scalar_net = BaseMLP(self.input_dim, hidden_dim, hidden_dim,
+                                     activation, last_act=True, flat=flat)
+        self.out_vector_net = BaseMLP(hidden_dim, hidden_dim,
+                                      n_vector_input * n_vector_input,
+                                      activation, flat=flat)
+
+    def forward(self, vectors_i, vectors_j, scalars=None):
+        z_i, z_j = vectors_i, vectors_j                       # [M, 3, K]
+        k = z_i.shape[-1]
+        s = torch.einsum("bij,bjk->bik", z_j.transpose(-1, -2),
+                         z_i).reshape(-1, k * k)
+        if self.norm:
+            s = F.normalize(s, p=2, dim=-1)
+        if scalars is not None:
+            s = torch.cat((s, scalars), dim=-1)
+        s = self.in_scalar_net(s)
+        vec_scalar = self.out_vector_net(s).reshape(-1, k, k)
+        vector = torch.einsum("bij,bjk->bik", z_j, vec_scalar)
+        return vector, s
+
+
+class PoolingLayer(nn.Module):
+    """EGHN pooling layer (reference basic.py:509-539)."""
+
+    def __init__(self, in_edge_nf, hidden_nf, n_vector_input,
+                 activation=None, flat=False):
+        super().__init__()
+        activation = activation if activation is not None else nn.SiLU()
+        self.edge_message_net = EquivariantEdgeScalarNet(
+            n_vector_input=n_vector_input, hidden_dim=hidden_nf,
+            activation=activation,
+            n_scalar_input=2 * hidden_nf + in_edge_nf, norm=True, flat=flat)
+        self.node_net = BaseMLP(2 * hidden_nf, hidden_nf, hidden_nf,
+                                activation, flat=flat)
+
+    def forward(self, vectors, h, edge_index, edge_fea):
+        row = edge_index[0]
+        hij = torch.cat((h[row], h[edge_index[1]], edge_fea), dim=-1)
+        vec_out, message = self.edge_message_net(
+            vectors_i=vectors[row], vectors_j=vectors[edge_index[1]],
+            scalars=hij)
+        dim, v = vec_out.shape[-2], vec_out.shape[-1]
+        agg_vec = aggregate(vec_out.reshape(-1, dim * v), row, h.shape[0],
+                            "mean").reshape(-1, dim, v)
+        vectors = vectors + agg_vec
+        tot = aggregate(message, row, h.shape[0], "sum")
+        h = self.node_net(torch.cat((h, tot), dim=-1)) + h
+        return vectors, h
+
+
+class PoolingNet(nn.Module):
+    """EGHN pooling network (reference basic.py:542-566)."""
+
+    def __init__(self, n_layers, in_edge_nf, n_vector_input, hidden_nf,
+                 output_nf, activation=None, device="cpu", flat=False):
+        super().__init__()
+        activation = activation if activation is not None else nn.SiLU()
+        self.layers = nn.ModuleList([
+            PoolingLayer(in_edge_nf, hidden_nf,
+                         n_vector_input=n_vector_input,
+                         activation=activation, flat=flat)
+            for _ in range(n_layers)])
+        self.n_layers = n_layers
+        self.pooling = nn.Sequential(nn.Linear(hidden_nf, 8 * hidden_nf),
+                                     nn.Tanh(),
+                                     nn.Linear(8 * hidden_nf, output_nf))
+
+    def forward(self, vectors, h, edge_index, edge_fea):
+        if isinstance(vectors, list):
+            vectors = torch.stack(vectors, dim=-1)
+        for layer in self.layers:
+            vectors, h = layer(vectors, h, edge_index, edge_fea)
+        return self.pooling(h)
+
+
+def _adj_matmul(edge_index, n, dense):
+    """(unweighted adjacency) @ dense — replaces torch_sparse.spmm
+    (reference basic.py:663,668): row i accumulates dense[col[e]]."""
+    msg = dense.index_select(0, edge_index[1])
+    return aggregate(msg, edge_index[0], n, "sum")
+
+
+class EGHN(nn.Module):
+    """Equivariant Graph Hierarchy Network (reference basic.py:569-731)."""
+
+    def __init__(self, in_node_nf, in_edge_nf, hidden_nf, n_cluster,
+                 layer_per_block=3, layer_pooling=3, layer_decoder=1,
+                 flat=False, activation=None, device="cpu", norm=False,
+                 with_v=True):
+        super().__init__()
+        activation = activation if activation is not None else nn.SiLU()
+        self.embedding = nn.Linear(in_node_nf, hidden_nf)
+        self.current_pooling_plan = None
+        self.n_cluster = n_cluster
+        self.with_v = with_v
+        self.low_force_net = EGNN(n_layers=layer_per_block,
+                                  in_node_nf=hidden_nf,
+                                  in_edge_nf=in_edge_nf, hidden_nf=hidden_nf,
+                                  activation=activation, with_v=with_v,
+                                  flat=flat, norm=norm)
+        self.low_pooling = PoolingNet(n_vector_input=3, hidden_nf=hidden_nf,
+                                      output_nf=n_cluster,
+                                      activation=activation,
+                                      in_edge_nf=in_edge_nf,
+                                      n_layers=layer_pooling, flat=flat)
+        self.high_force_net = EGNN(n_layers=layer_per_block,
+                                   in_node_nf=hidden_nf, in_edge_nf=1,
+                                   hidden_nf=hidden_nf,
+                                   activation=activation, with_v=with_v,
+                                   flat=flat)
+        nvi = 4 if with_v else 3
+        if layer_decoder == 1:
+            self.kinematics_net = EquivariantScalarNet(
+                n_vector_input=nvi, hidden_dim=hidden_nf,
+                activation=activation, n_scalar_input=2 * hidden_nf,
+                norm=True, flat=flat)
+        else:
+            self.kinematics_net = EGMN(
+                n_vector_input=nvi, hidden_dim=hidden_nf,
+                activation=activation, n_scalar_input=2 * hidden_nf,
+                norm=True, flat=flat, n_layers=layer_decoder)
+
+    def get_cut_loss(self, a):
+        a = F.normalize(a, p=2, dim=2)
+        eye = torch.eye(a.shape[-1], device=a.device)
+        return torch.norm(a - eye, p="fro", dim=[1, 2]).mean()
+
+    @staticmethod
+    def construct_edges(a, n_node):
+        h_edge_fea = a.reshape(-1)
+        p = a.shape[1]
+        h_row = torch.arange(p, device=a.device).unsqueeze(-1) \
+            .expand(-1, p).reshape(-1)
+        h_col = torch.arange(p, device=a.device).unsqueeze(0) \
+            .expand(p, -1).reshape(-1)
+        h_row = h_row.unsqueeze(0).expand(a.shape[0], -1)
+        h_col = h_col.unsqueeze(0).expand(a.shape[0], -1)
+        offset = (torch.arange(a.shape[0], device=a.device)
+                  * n_node).unsqueeze(-1)
+        h_row = (h_row + offset).reshape(-1)
+        h_col = (h_col + offset).reshape(-1)
+        h_edge_mask = torch.ones_like(h_row)
+        h_edge_mask[torch.arange(p, device=a.device) * (p + 1)] = 0
+        return h_row, h_col, h_edge_fea, h_edge_mask
+
+    def forward(self, x, h, edge_index, edge_fea, local_edge_index,
+                local_edge_fea, n_node, v=None, node_mask=None,
+                node_nums=None):
+        h = self.embedding(h)
+        row, col = edge_index[0], edge_index[1]
+
+        new_x, new_v, h = self.low_force_net(x, h, edge_index, edge_fea, v=v)
+        nf = new_x - x
+
+        if node_nums is None:
+            x_mean = x.reshape(-1, n_node, x.shape[-1]).mean(
+                1, keepdim=True).expand(-1, n_node, -1).reshape(
+                -1, x.shape[-1])
+        else:
+            pooled = (x.reshape(-1, n_node, x.shape[-1]).sum(1).T
+                      / node_nums).T.unsqueeze(1)
+            x_mean = pooled.expand(-1, n_node, -1).reshape(-1, x.shape[-1])
+        pooling_fea = self.low_pooling(vectors=[x - x_mean, nf, v], h=h,
+                                       edge_index=local_edge_index,
+                                       edge_fea=local_edge_fea)
+        hard = F.one_hot(pooling_fea.argmax(-1), self.n_cluster).float()
+        pooling = F.softmax(pooling_fea, dim=1)
+        self.current_pooling_plan = hard
+
+        s = pooling.reshape(-1, n_node, pooling.shape[-1])     # [B, N, P]
+        s_t = s.transpose(-2, -1)
+        p_index = torch.ones_like(nf)[..., 0]
+        if node_mask is not None:
+            p_index = p_index * node_mask
+        p_index = p_index.reshape(-1, n_node, 1)
+        count = torch.einsum("bij,bjk->bik", s_t, p_index).clamp_min(1e-5)
+        _x = x.reshape(-1, n_node, x.shape[-1])
+        _h = h.reshape(-1, n_node, h.shape[-1])
+        _nf = nf.reshape(-1, n_node, nf.shape[-1])
+        big_x = torch.einsum("bij,bjk->bik", s_t, _x) / count
+        big_h = torch.einsum("bij,bjk->bik", s_t, _h) / count
+        big_nf = torch.einsum("bij,bjk->bik", s_t, _nf) / count
+        if v is not None:
+            big_v = torch.einsum("bij,bjk->bik", s_t,
+                                 v.reshape(-1, n_node, v.shape[-1])) / count
+            big_v = big_v.reshape(-1, big_v.shape[-1])
+        else:
+            big_v = None
+        big_x = big_x.reshape(-1, big_x.shape[-1])
+        big_h = big_h.reshape(-1, big_h.shape[-1])
+
+        a = _adj_matmul(local_edge_index, x.shape[0], pooling)
+        a = a.reshape(-1, n_node, a.shape[-1])
+        big_a = torch.einsum("bij,bjk->bik", s_t, a)
+        self.cut_loss = self.get_cut_loss(big_a)
+        aa = _adj_matmul(edge_index, x.shape[0], pooling)
+        aa = aa.reshape(-1, n_node, aa.shape[-1])
+        big_aa = torch.einsum("bij,bjk->bik", s_t, aa)
+
+        h_row, h_col, h_edge_fea, _ = self.construct_edges(
+            big_aa, big_aa.shape[-1])
+        h_new_x, h_new_v, h_new_h = self.high_force_net(
+            big_x, big_h, (h_row, h_col), h_edge_fea.unsqueeze(-1), v=big_v)
+        h_nf = h_new_x - big_x
+
+        p = big_aa.shape[1]
+        l_nf = torch.einsum("bij,bjk->bik", s,
+                            h_nf.reshape(-1, p, 3)).reshape(-1, 3)
+        l_x = torch.einsum("bij,bjk->bik", s,
+                           big_x.reshape(-1, p, 3)).reshape(-1, 3)
+        if v is not None:
+            l_v = torch.einsum("bij,bjk->bik", s,
+                               big_v.reshape(-1, p, 3)).reshape(-1, 3)
+            vectors = [l_nf, x - l_x, v - l_v, nf]
+        else:
+            vectors = [l_nf, x - l_x, nf]
+        l_h = torch.einsum("bij,bjk->bik", s,
+                           h_new_h.reshape(-1, p, h_new_h.shape[-1]))
+        l_h = l_h.reshape(-1, l_h.shape[-1])
+        l_kin, h_out = self.kinematics_net(
+            vectors=vectors, scalars=torch.cat((h, l_h), dim=-1))
+        _l_x = torch.einsum("bij,bjk->bik", s,
+                            (big_x + h_nf).reshape(-1, p, 3)).reshape(-1, 3)
+        x_out = _l_x + l_kin
+        return (x_out, v, h_out) if v is not None else (x_out, h_out)
+
+
+class FullMLP(nn.ModuleList):
+    """Non-equivariant MLP baseline (reference basic.py:734-749)."""
+
+    def __init__(self, in_node_nf, hidden_nf, n_layers, activation=None,
+                 flat=False, device="cpu"):
+        super().__init__()
+        activation = activation if activation is not None else nn.SiLU()
+        self.layers = nn.ModuleList([
+            BaseMLP(hidden_nf, hidden_nf, hidden_nf, activation,
+                    residual=True, last_act=True, flat=flat)
+            for _ in range(n_layers)])
+        self.embedding = nn.Linear(in_node_nf, hidden_nf)
+        self.output = nn.Linear(hidden_nf, 3)
+
+    def forward(self, x):
+        x = self.embedding(x)
+        for layer in self.layers:
+            x = layer(x)
+        return self.output(x)

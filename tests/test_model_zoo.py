@@ -152,3 +152,27 @@ def test_se3_transformer_forward():
     out = m(b.pos, b.vel, b.attr, b.edge_index)
     assert out.shape == (b.num_nodes, 3)
     assert torch.isfinite(out).all()
+
+
+def test_eghn_forward_and_equivariance():
+    from distegnn_amd.models.baselines import EGHN
+
+    fix_seed(8)
+    b = batch(2, seed=10)
+    m = EGHN(in_node_nf=2, in_edge_nf=2, hidden_nf=8, n_cluster=3,
+             layer_per_block=1, layer_pooling=1).double()
+    n_node = 100
+
+    def run(pos, vel):
+        x_out, _, _ = m(pos, b.x.double(), b.edge_index,
+                        b.edge_attr.double(), b.edge_index,
+                        b.edge_attr.double(), n_node, v=vel)
+        return x_out
+
+    o1 = run(b.pos.double(), b.vel.double())
+    assert o1.shape == (b.num_nodes, 3)
+    R = torch.tensor(rotate.random_rotate(np.random.default_rng(11)))
+    o2 = run(b.pos.double() @ R, b.vel.double() @ R)
+    # EGHN is O(3)-equivariant around the per-graph mean (translation is
+    # removed internally via x_mean); test rotation equivariance
+    assert torch.allclose(o1 @ R, o2, atol=1e-7), (o1 @ R - o2).abs().max()
