@@ -147,12 +147,6 @@ class EGCLVel(nn.Module):
         row = edge_index[0]
         dist_active = self.world_size > 1 and comm.is_distributed()
 
-        # (X_c - x_i): [N, C, 3]; its norm: [N, C, 1]
-        vdiff = (ops.gather_rows(virtual_coord, batch, ptr,
-                                 chunks=pool_chunks)
-                 - coord.unsqueeze(1))
-        vradial = vdiff.norm(p=2, dim=-1, keepdim=True)
-
         # --- edge block (real-real): phi_e messages + phi_x translations -
         # Fusable form (the standard FastEGNN configuration): one MFMA HIP
         # kernel computes gather + phi_e MLP + phi_x head + d_ij * phi_x and
@@ -194,16 +188,44 @@ class EGCLVel(nn.Module):
         m_x = virtual_coord - coord_mean.unsqueeze(1)
         gram = torch.matmul(m_x, m_x.transpose(1, 2))
 
-        # --- virtual edge model, phi_ev: [N, C, 2H+1+C] -> [N, C, H] ----
-        v_in = torch.cat([
-            h.unsqueeze(1).expand(n, c, h.size(1)),
-            ops.gather_rows(virtual_feat, batch, ptr, chunks=pool_chunks),
-            vradial,
-            ops.gather_rows(gram, batch, ptr, chunks=pool_chunks),
-        ], dim=-1)
-        v_msg = self.edge_mlp_virtual(v_in)
-        if self.attention:
-            v_msg = v_msg * self.att_mlp_virtual(v_msg)
+        # --- virtual edge block, phi_ev + phi_xv/phi_X heads ------------
+        # Fusable form: one MFMA kernel over (node, channel) rows builds
+        # the inputs (vdiff / vradial / concat) in LDS and returns per-row
+        # messages + head translations (ops.fused_virtual_block;
+        # csrc/fused_virtual.hip). attention/tanh variants compose eagerly.
+        vfuse = not self.attention and not self.tanh
+        if vfuse:
+            v_msg, tv, trans_x = ops.fused_virtual_block(
+                h, coord, virtual_coord, virtual_feat, gram, batch, ptr,
+                pool_chunks,
+                self.edge_mlp_virtual[0].weight,
+                self.edge_mlp_virtual[0].bias,
+                self.edge_mlp_virtual[2].weight,
+                self.edge_mlp_virtual[2].bias,
+                self.coord_mlp_r_virtual[0].weight,
+                self.coord_mlp_r_virtual[0].bias,
+                self.coord_mlp_r_virtual[2].weight.reshape(-1),
+                self.coord_mlp_v_virtual[0].weight,
+                self.coord_mlp_v_virtual[0].bias,
+                self.coord_mlp_v_virtual[2].weight.reshape(-1))
+            trans_v = tv.mean(dim=1)
+        else:
+            vdiff = (ops.gather_rows(virtual_coord, batch, ptr,
+                                     chunks=pool_chunks)
+                     - coord.unsqueeze(1))
+            vradial = vdiff.norm(p=2, dim=-1, keepdim=True)
+            v_in = torch.cat([
+                h.unsqueeze(1).expand(n, c, h.size(1)),
+                ops.gather_rows(virtual_feat, batch, ptr,
+                                chunks=pool_chunks),
+                vradial,
+                ops.gather_rows(gram, batch, ptr, chunks=pool_chunks),
+            ], dim=-1)
+            v_msg = self.edge_mlp_virtual(v_in)
+            if self.attention:
+                v_msg = v_msg * self.att_mlp_virtual(v_msg)
+            trans_v = (-vdiff * self.coord_mlp_r_virtual(v_msg)).mean(dim=1)
+            trans_x = vdiff * self.coord_mlp_v_virtual(v_msg)  # [N, C, 3]
 
         # --- coord model (real), phi_x / phi_xv / phi_v ------------------
         if not fuse:
@@ -215,14 +237,12 @@ class EGCLVel(nn.Module):
             else:
                 raise ValueError(f"coords_agg {self.coords_agg}")
         coord = coord + agg
-        trans_v = (-vdiff * self.coord_mlp_r_virtual(v_msg)).mean(dim=1)
         coord = coord + trans_v
         coord = coord + self.coord_mlp_vel(h) * vel
         if self.gravity is not None:
             coord = coord + self.gravity_mlp(h) * self.gravity.to(h.device)
 
         # --- virtual aggregates (fused site B+C collective) -------------
-        trans_x = vdiff * self.coord_mlp_v_virtual(v_msg)      # [N, C, 3]
         agg_vc = ops.graph_mean_pool(
             trans_x.reshape(n, -1), batch, b, ptr=ptr, counts=counts,
             chunks=pool_chunks

@@ -339,6 +339,141 @@ def fused_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
             segment_mean(trans, row, n, rowptr=rowptr))
 
 
+def eager_virtual_block(h, coord, vcoord, vfeat, gram, batch,
+                        w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv):
+    """Eager composition of the fused virtual block (same math).
+
+    Returns (v_msg [N,C,H], tv [N,C,3], tx [N,C,3])."""
+    import torch.nn.functional as F
+
+    n = h.size(0)
+    c = vcoord.size(1)
+    dt = h.dtype
+    w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv = (
+        t.to(dt) for t in (w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv))
+    vdiff = vcoord.index_select(0, batch) - coord.unsqueeze(1)  # [N,C,3]
+    vrad = vdiff.norm(p=2, dim=-1, keepdim=True)
+    v_in = torch.cat([
+        h.unsqueeze(1).expand(n, c, h.size(1)),
+        vfeat.index_select(0, batch).to(dt),
+        vrad.to(dt),
+        gram.index_select(0, batch).to(dt),
+    ], dim=-1)
+    t1 = F.silu(F.linear(v_in, w1, b1))
+    vmsg = F.silu(F.linear(t1, w2, b2))
+    pxv = (F.silu(F.linear(vmsg, wxv, bxv)) @ wxvv).unsqueeze(-1).float()
+    p_x = (F.silu(F.linear(vmsg, wX, bX)) @ wXv).unsqueeze(-1).float()
+    tv = -vdiff * pxv
+    tx = vdiff * p_x
+    return vmsg, tv, tx
+
+
+class _FusedVirtualBlockFn(torch.autograd.Function):
+    """Fused MFMA virtual-edge block (csrc/fused_virtual.hip).
+
+    Forward: one kernel over (node, channel) rows builds the virtual-edge
+    inputs in LDS (no [N,C,3]/[N,C,134] torch intermediates) and returns
+    per-row messages + head translations; training mode saves the
+    pre-activations. Backward: one kernel runs the dz chain from the saved
+    pre-activations; python finishes with split-K wgrad GEMMs, bias sums,
+    a sum-over-channels for dh/dcoord and per-graph pools for
+    dvfeat/dgram/dvcoord."""
+
+    @staticmethod
+    def forward(ctx, h, coord, vcoord, vfeat, gram, batch, ptr, chunks_cb,
+                chunks_ce, chunks_scp, w1, b1, w2, b2, wxv, bxv, wxvv, wX,
+                bX, wXv, train):
+        ext = _require_ext("fused_virtual_block")
+        outs = ext.fused_virtual_forward(
+            h, coord, vcoord.float(), vfeat.bfloat16(), gram.float(), batch,
+            w1.bfloat16(), b1, w2.bfloat16(), b2, wxv.bfloat16(), bxv, wxvv,
+            wX.bfloat16(), bX, wXv, train)
+        vmsg, tv, tx, vin, z1, z2, zxv, zX, p2 = outs
+        n, c = h.size(0), vcoord.size(1)
+        ctx.save_for_backward(h, coord, vcoord, batch, ptr, chunks_cb,
+                              chunks_ce, chunks_scp, vin, z1, z2, zxv, zX,
+                              p2, vmsg, w1, w2, wxv, wX, wxvv, wXv)
+        ctx.shape_nc = (n, c)
+        ctx.vfeat_dtype = vfeat.dtype
+        return (vmsg.view(n, c, -1), tv.view(n, c, 3), tx.view(n, c, 3))
+
+    @staticmethod
+    def backward(ctx, dvmsg, dtv, dtx):
+        (h, coord, vcoord, batch, ptr, cb, ce, scp, vin, z1, z2, zxv, zX,
+         p2, vmsg, w1, w2, wxv, wX, wxvv, wXv) = ctx.saved_tensors
+        ext = _load_extension()
+        n, c = ctx.shape_nc
+        rows = n * c
+        (dz1, dz2, dzxv, dzX, dh_row, dvf_row, dgram_row, dvd,
+         dp2) = ext.fused_virtual_backward(
+            coord, vcoord.float(), batch,
+            dvmsg.reshape(rows, -1).to(torch.bfloat16).contiguous(),
+            dtv.reshape(rows, 3).float().contiguous(),
+            dtx.reshape(rows, 3).float().contiguous(),
+            z1, z2, zxv, zX, p2, w1.bfloat16(), w2.bfloat16(),
+            wxv.bfloat16(), wX.bfloat16(), wxvv, wXv)
+        from .linear import chunked_wgrad
+
+        k_in = w1.size(1)
+        t1 = torch.nn.functional.silu(z1)
+        gw1 = chunked_wgrad(dz1, vin)[:, :k_in].float()
+        gw2 = chunked_wgrad(dz2, t1).float()
+        gwxv = chunked_wgrad(dzxv, vmsg).float()
+        gwX = chunked_wgrad(dzX, vmsg).float()
+        gb1 = dz1.sum(0, dtype=torch.float32)
+        gb2 = dz2.sum(0, dtype=torch.float32)
+        gbxv = dzxv.sum(0, dtype=torch.float32)
+        gbX = dzX.sum(0, dtype=torch.float32)
+        sxv = torch.nn.functional.silu(zxv)
+        s_x = torch.nn.functional.silu(zX)
+        gwxvv = (sxv.float() * dp2[:, :1]).sum(0)
+        gwXv = (s_x.float() * dp2[:, 1:]).sum(0)
+
+        gh = dh_row.view(n, c, -1).sum(1, dtype=torch.float32).to(h.dtype)
+        gcoord = -dvd.view(n, c, 3).sum(1)
+        # per-graph pools (sum) of the [B,C,*] gradients
+        def pool(x_rows, width):
+            flat = x_rows.reshape(n, c * width)
+            if cb.numel():
+                out = ext.segment_reduce_chunked(flat.contiguous(), ptr, cb,
+                                                 ce, scp, False)
+            else:
+                out = ext.segment_reduce_csr(flat.contiguous(), ptr, False)
+            return out.view(-1, c, width)
+
+        gvfeat = pool(dvf_row, dvf_row.size(1)).to(ctx.vfeat_dtype)
+        ggram = pool(dgram_row[:, :c].contiguous(), c).float()
+        gvcoord = pool(dvd, 3)
+        return (gh, gcoord, gvcoord, gvfeat, ggram, None, None, None, None,
+                None, gw1, gb1, gw2, gb2, gwxv, gbxv, gwxvv, gwX, gbX, gwXv,
+                None)
+
+
+def fused_virtual_block(h, coord, vcoord, vfeat, gram, batch, ptr, chunks,
+                        w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv):
+    """Dispatch: HIP fused kernel on GPU bf16 H=64 C<=8, eager otherwise.
+
+    Returns (v_msg [N,C,H], tv [N,C,3], tx [N,C,3]); the model derives
+    trans_v = tv.mean(1), agg_v = v_msg.mean(1), and pools tx / v_msg."""
+    usable = (h.is_cuda and h.dtype == torch.bfloat16 and h.size(1) == 64
+              and vcoord.size(1) <= 8 and ptr is not None
+              and hip_ext() is not None
+              and os.environ.get("DISTEGNN_DISABLE_FUSED") != "1")
+    if usable:
+        cb, ce, scp = chunks if chunks is not None else (None, None, None)
+        empty = batch.new_empty(0)
+        train = torch.is_grad_enabled() and (
+            h.requires_grad or w1.requires_grad or coord.requires_grad)
+        return _FusedVirtualBlockFn.apply(
+            h, coord, vcoord, vfeat, gram, batch, ptr,
+            cb if cb is not None else empty,
+            ce if ce is not None else empty,
+            scp if scp is not None else empty,
+            w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv, train)
+    return eager_virtual_block(h, coord, vcoord, vfeat, gram, batch,
+                               w1, b1, w2, b2, wxv, bxv, wxvv, wX, bX, wXv)
+
+
 def radius_graph(pos: torch.Tensor, r: float, loop: bool = False) -> torch.Tensor:
     """Directed radius graph, row-sorted. GPU: HIP cell-list kernel."""
     if pos.is_cuda and r is not None and r >= 0:
@@ -350,6 +485,7 @@ def radius_graph(pos: torch.Tensor, r: float, loop: bool = False) -> torch.Tenso
 
 __all__ = [
     "segment_sum", "segment_mean", "graph_sum_pool", "graph_mean_pool",
-    "gather_rows", "fused_edge_block", "eager_edge_block", "radius_graph",
+    "gather_rows", "fused_edge_block", "eager_edge_block",
+    "fused_virtual_block", "eager_virtual_block", "radius_graph",
     "hip_ext", "reference",
 ]

@@ -25,6 +25,7 @@ SOURCES = [
     os.path.join(_CSRC, "fused_edge_bwd.hip"),
     os.path.join(_CSRC, "wgrad.hip"),
     os.path.join(_CSRC, "tall_linear.hip"),
+    os.path.join(_CSRC, "fused_virtual.hip"),
 ]
 
 

@@ -18,6 +18,18 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
 torch::Tensor tall_linear(torch::Tensor x, torch::Tensor bmat,
                           c10::optional<torch::Tensor> bias, int64_t act);
+std::vector<torch::Tensor> fused_virtual_forward(
+    torch::Tensor h, torch::Tensor coord, torch::Tensor vcoord,
+    torch::Tensor vfeat, torch::Tensor gram, torch::Tensor batch,
+    torch::Tensor w1, torch::Tensor b1, torch::Tensor w2, torch::Tensor b2,
+    torch::Tensor wxv, torch::Tensor bxv, torch::Tensor wxvv,
+    torch::Tensor wX, torch::Tensor bX, torch::Tensor wXv, bool train);
+std::vector<torch::Tensor> fused_virtual_backward(
+    torch::Tensor coord, torch::Tensor vcoord, torch::Tensor batch,
+    torch::Tensor dvmsg, torch::Tensor dtv, torch::Tensor dtx,
+    torch::Tensor z1, torch::Tensor z2, torch::Tensor zxv, torch::Tensor zX,
+    torch::Tensor p2, torch::Tensor w1, torch::Tensor w2, torch::Tensor wxv,
+    torch::Tensor wX, torch::Tensor wxvv, torch::Tensor wXv);
 std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
@@ -63,6 +75,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "tall-skinny MFMA linear: act(x @ B^T + bias)",
         py::arg("x"), py::arg("bmat"), py::arg("bias") = c10::nullopt,
         py::arg("act") = 0);
+  m.def("fused_virtual_forward", &fused_virtual_forward,
+        "fused MFMA virtual-edge block forward");
+  m.def("fused_virtual_backward", &fused_virtual_backward,
+        "fused MFMA virtual-edge block backward");
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
         py::arg("a"), py::arg("bt"));
