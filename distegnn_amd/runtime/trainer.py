@@ -34,6 +34,7 @@ from torch import nn
 
 from ..parallel import comm
 from ..parallel.comm import GradBucket
+from .graphs import GraphedStep
 from .losses import mmd_loss
 
 try:
@@ -104,11 +105,42 @@ def _is_fast_model(model_name: str) -> bool:
     return model_name.startswith("Fast")
 
 
+def make_train_step_core(model, model_name, loss_fn, train_config,
+                         autocast_dtype, device):
+    """forward + loss (+MMD) + accumulation-scaled backward on
+    device-resident batch tensors — hipGraph-capturable (no host syncs,
+    no collectives; counts_global is reduced eagerly by the caller)."""
+
+    def step_core(data):
+        total_node_cnt = data.counts_global.sum()
+        node_cnt = float(data.num_nodes)
+        with contextlib.ExitStack() as stack:
+            if autocast_dtype is not None:
+                stack.enter_context(torch.autocast("cuda",
+                                                   dtype=autocast_dtype))
+            loc_pred, virtual_node_loc = model_forward(
+                model, model_name, data, device)
+        loss_loc = loss_fn(loc_pred.float(), data.target)
+        weight = node_cnt / total_node_cnt
+        loss_loc = weight * loss_loc
+        mse_log = loss_loc.detach()
+        if _is_fast_model(model_name) and virtual_node_loc is not None:
+            vloc = virtual_node_loc.permute(0, 2, 1).float()
+            lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
+                          data.counts, train_config.mmd.sigma,
+                          train_config.mmd.samples)
+            loss_loc = loss_loc + train_config.mmd.weight * weight * lm
+        (loss_loc / float(train_config.accumulation_steps)).backward()
+        return (mse_log,)
+
+    return step_core
+
+
 def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
                        loss_fn, dataset_name, train_config, epoch_index, tag,
                        subgraphs, world_size, device, grad_bucket=None,
                        autocast_dtype=None, debug_lockstep=False,
-                       progress=True, step_timer=None):
+                       progress=True, step_timer=None, graphed_step=None):
     backprop = tag == "train"
     if backprop:
         model.train()
@@ -145,34 +177,41 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
         # one collective per step: per-graph global node counts
         data.counts_global = comm.global_counts(data.counts) \
             if world_size > 1 else data.counts
-        total_node_cnt = data.counts_global.sum()
-        node_cnt = float(data.num_nodes)
 
-        with contextlib.ExitStack() as stack:
-            if not backprop:
-                stack.enter_context(torch.no_grad())
-            if autocast_dtype is not None:
-                stack.enter_context(torch.autocast("cuda",
-                                                   dtype=autocast_dtype))
-            loc_pred, virtual_node_loc = model_forward(
-                model, model_name, data, device)
+        if backprop and graphed_step is not None:
+            (mse_log,) = graphed_step(data)
+            loss_accum = loss_accum + mse_log * batch_size
+            counter = counter + batch_size
+        else:
+            total_node_cnt = data.counts_global.sum()
+            node_cnt = float(data.num_nodes)
+            with contextlib.ExitStack() as stack:
+                if not backprop:
+                    stack.enter_context(torch.no_grad())
+                if autocast_dtype is not None:
+                    stack.enter_context(torch.autocast("cuda",
+                                                       dtype=autocast_dtype))
+                loc_pred, virtual_node_loc = model_forward(
+                    model, model_name, data, device)
 
-        loss_loc = loss_fn(loc_pred.float(), data.target)
-        # node-count weighting: per-rank share of the global per-node MSE
-        weight = node_cnt / total_node_cnt
-        loss_loc = weight * loss_loc
-        loss_accum = loss_accum + loss_loc.detach() * batch_size
-        counter = counter + batch_size
+            loss_loc = loss_fn(loc_pred.float(), data.target)
+            # node-count weighting: rank share of the global per-node MSE
+            weight = node_cnt / total_node_cnt
+            loss_loc = weight * loss_loc
+            loss_accum = loss_accum + loss_loc.detach() * batch_size
+            counter = counter + batch_size
 
-        if _is_fast_model(model_name) and virtual_node_loc is not None:
-            vloc = virtual_node_loc.permute(0, 2, 1).float()  # [B, C, 3]
-            lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
-                          data.counts, train_config.mmd.sigma,
-                          train_config.mmd.samples)
-            loss_loc = loss_loc + train_config.mmd.weight * weight * lm
+            if _is_fast_model(model_name) and virtual_node_loc is not None:
+                vloc = virtual_node_loc.permute(0, 2, 1).float()  # [B,C,3]
+                lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
+                              data.counts, train_config.mmd.sigma,
+                              train_config.mmd.samples)
+                loss_loc = loss_loc + train_config.mmd.weight * weight * lm
+            if backprop:
+                (loss_loc
+                 / float(train_config.accumulation_steps)).backward()
 
         if backprop:
-            (loss_loc / float(train_config.accumulation_steps)).backward()
             if (step + 1) % train_config.accumulation_steps == 0:
                 if grad_bucket is not None:
                     grad_bucket.sync()              # SUM == ref avg * ws
@@ -208,6 +247,19 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
     debug_lockstep = bool(train_config.get("debug_lockstep", False) or
                           os.environ.get("DISTEGNN_DEBUG_LOCKSTEP") == "1")
 
+    # hipGraph-captured train step (config train.hip_graphs: auto|on|off;
+    # auto = on for single-GPU CUDA runs)
+    hg = str(train_config.get("hip_graphs", "auto")).lower()
+    use_graphs = (hg in ("on", "true") or
+                  (hg == "auto" and world_size == 1)) \
+        and device.type == "cuda"
+    graphed_step = None
+    if use_graphs:
+        graphed_step = GraphedStep(
+            make_train_step_core(model, model_name, loss_mse, train_config,
+                                 autocast_dtype, device),
+            model.parameters(), warmup_occurrences=2)
+
     log_dict = {"epochs": [], "loss": [], "loss_train": []}
     best_log_dict = {"epoch_index": 0, "loss_valid": 1e8, "loss_test": 1e8,
                      "loss_train": 1e8}
@@ -232,7 +284,7 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
             tag="train", subgraphs=config.model.virtual_channels,
             world_size=world_size, device=device, grad_bucket=grad_bucket,
             autocast_dtype=autocast_dtype, debug_lockstep=debug_lockstep,
-            progress=progress)
+            progress=progress, graphed_step=graphed_step)
         if rank == 0:
             log_dict["loss_train"].append(loss_train)
 
