@@ -228,6 +228,11 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
     # deferred logging reduce: one collective per epoch (ref: per step)
     if world_size > 1:
         dist.all_reduce(loss_accum, op=dist.ReduceOp.SUM)
+    if counter.item() == 0:
+        if rank == 0:
+            print(f"WARNING: {tag} loader produced no batches (dataset "
+                  f"smaller than batch_size with drop_last) — skipping")
+        return float("nan")
     avg = (loss_accum / counter).item()
     if rank == 0:
         prefix = "" if backprop else "==> "
