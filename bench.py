@@ -26,7 +26,7 @@ import time
 import torch
 
 from distegnn_amd.data.graph import collate
-from distegnn_amd.data.synthetic import make_cloud_sample
+from distegnn_amd.data.synthetic import make_cloud_sample, make_cutoff_dataset
 from distegnn_amd.data.partition import SPLITTERS
 from distegnn_amd.models import FastEGNN
 from distegnn_amd.parallel import comm
@@ -97,7 +97,7 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
 
 
 def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
-               world_size, device):
+               world_size, device, clip=True):
     data = batch.to(device)
     data.counts_global = (comm.global_counts(data.counts)
                           if world_size > 1 else data.counts)
@@ -105,7 +105,9 @@ def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
     if (step + 1) % accum == 0:
         if grad_bucket is not None:
             grad_bucket.sync()
-        torch.nn.utils.clip_grad_norm_(model_params(graphed), max_norm=0.3)
+        if clip:  # reference clip rule: FastEGNN + (ws>1 or LargeFluid)
+            torch.nn.utils.clip_grad_norm_(model_params(graphed),
+                                           max_norm=0.3)
         optimizer.step()
         optimizer.zero_grad(set_to_none=False)
     return mse_log
@@ -115,18 +117,40 @@ def model_params(graphed):
     return graphed.params
 
 
+# BASELINE.md workloads (reference dataset scales; see data/synthetic.py)
+WORKLOADS = {
+    # name: (dataset, nodes/graph, graphs/batch, radius, feat, attr, C,
+    #        accum, mmd_sigma, mmd_samples, normalize)
+    "largefluid": ("Fluid113K", 113140, 1, 0.075, 3, 2, 5, 4, 3.0, 50, False),
+    "water3d": ("Water-3D", 7806, 15, 0.035, 2, 0, 3, 1, 1.5, 3, False),
+    "protein": ("protein", 855, 5, 10.0, 2, 0, 3, 1, 1.0, 3, False),
+    "nbody": ("nbody_100", 100, 250, -1.0, 2, 0, 3, 1, 1.5, 3, True),
+}
+
+
+def build_cutoff_batches(workload, num_batches, graphs_per_batch, seed):
+    """Single-device ("cutoff_edges") batches at published scales."""
+    dataset, nodes, _, radius, *_ = WORKLOADS[workload]
+    samples = make_cutoff_dataset(dataset, num_batches * graphs_per_batch,
+                                  seed=seed)
+    return [collate(samples[i * graphs_per_batch:(i + 1) * graphs_per_batch])
+            for i in range(num_batches)]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=16)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--nodes", type=int, default=113140)
-    ap.add_argument("--radius", type=float, default=0.075)
+    ap.add_argument("--workload", type=str, default="largefluid",
+                    choices=list(WORKLOADS))
+    ap.add_argument("--nodes", type=int, default=None)
+    ap.add_argument("--radius", type=float, default=None)
     ap.add_argument("--split-mode", type=str, default="random",
                     choices=["random", "metis", "kmeans"])
     ap.add_argument("--dtype", type=str, default="bf16",
                     choices=["bf16", "fp32"])
-    ap.add_argument("--virtual-channels", type=int, default=5)
+    ap.add_argument("--virtual-channels", type=int, default=None)
     ap.add_argument("--num-batches", type=int, default=2,
                     help="distinct synthetic samples to cycle through")
     ap.add_argument("--graphs", type=str, default="auto",
@@ -145,19 +169,32 @@ def main():
     device = torch.device(f"cuda:{rank}")
     torch.cuda.set_device(device)
 
+    (dataset, wl_nodes, graphs_per_batch, wl_radius, feat_nf, attr_nf, wl_c,
+     accum, mmd_sigma, mmd_samples, normalize) = WORKLOADS[args.workload]
+    nodes = args.nodes if args.nodes is not None else wl_nodes
+    radius = args.radius if args.radius is not None else wl_radius
+    vch = (args.virtual_channels if args.virtual_channels is not None
+           else wl_c)
+
     fix_seed(43)
     t_data = time.perf_counter()
-    batches = build_rank_batches(rank, world_size, args.num_batches,
-                                 args.nodes, args.radius, args.split_mode,
-                                 seed=43)
+    if args.workload == "largefluid":
+        batches = build_rank_batches(rank, world_size, args.num_batches,
+                                     nodes, radius, args.split_mode,
+                                     seed=43)
+    else:
+        assert world_size == 1,             "cutoff-mode workloads are single-device (reference main.py:173)"
+        batches = build_cutoff_batches(args.workload, args.num_batches,
+                                       graphs_per_batch, seed=43)
     if rank == 0:
         print(f"# data built in {time.perf_counter() - t_data:.1f}s: "
               f"{batches[0].num_nodes} nodes/rank, "
               f"{batches[0].num_edges} edges/rank", flush=True)
 
-    model = FastEGNN(node_feat_nf=3, node_attr_nf=2, edge_attr_nf=2,
-                     hidden_nf=64, virtual_channels=args.virtual_channels,
-                     world_size=world_size, n_layers=4).to(device)
+    model = FastEGNN(node_feat_nf=feat_nf, node_attr_nf=attr_nf,
+                     edge_attr_nf=2, hidden_nf=64, virtual_channels=vch,
+                     world_size=world_size, n_layers=4,
+                     normalize=normalize).to(device)
     grad_bucket = None
     if world_size > 1:
         grad_bucket = GradBucket(model)
@@ -165,8 +202,6 @@ def main():
     optimizer = torch.optim.Adam(model.parameters(), lr=5e-4,
                                  weight_decay=1e-12)
     autocast_dtype = torch.bfloat16 if args.dtype == "bf16" else None
-    accum = 4
-    mmd_sigma, mmd_samples = 3.0, 50
     model.train()
 
     from distegnn_amd.runtime.graphs import GraphedStep
@@ -179,17 +214,18 @@ def main():
                           warmup_occurrences=2,
                           enabled=use_graphs, verbose=True)
 
+    clip = args.workload == "largefluid" or world_size > 1
     mse = None
     for w in range(args.warmup):
         mse = train_step(graphed, batches[w % len(batches)], optimizer,
-                         grad_bucket, w, accum, world_size, device)
+                         grad_bucket, w, accum, world_size, device, clip)
     comm.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for k in range(args.steps):
         mse = train_step(graphed, batches[k % len(batches)], optimizer,
                          grad_bucket, args.warmup + k, accum,
-                         world_size, device)
+                         world_size, device, clip)
     comm.barrier()
     torch.cuda.synchronize()
     elapsed = torch.tensor(time.perf_counter() - t0, device=device)
@@ -213,15 +249,16 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "model": "FastEGNN-DistEGNN",
-                "dataset": "LargeFluid-113K",
-                "global_batch": 1,
-                "nodes": args.nodes,
-                "radius": args.radius,
+                "model": ("FastEGNN-DistEGNN" if args.workload ==
+                          "largefluid" else "FastEGNN"),
+                "dataset": dataset,
+                "global_batch": graphs_per_batch,
+                "nodes": nodes,
+                "radius": radius,
                 "split_mode": args.split_mode,
                 "hidden_nf": 64,
                 "n_layers": 4,
-                "virtual_channels": args.virtual_channels,
+                "virtual_channels": vch,
                 "accumulation_steps": accum,
                 "parallelism": f"graph-partition dp{world_size}",
                 "coord_mse": float(mse.item()) if mse is not None else None,
