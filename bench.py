@@ -79,7 +79,8 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
         with ctx:
             loc_pred, vloc = model(
                 data.x, data.pos, data.vel, data.loc_mean, data.edge_index,
-                data.batch, edge_attr=data.edge_attr, node_attr=data.attr,
+                data.batch, edge_attr=data.edge_attr,
+                node_attr=(data.attr if model.node_attr_nf else None),
                 rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
                 counts_global=data.counts_global, pool_chunks=chunks,
                 colptr=data.colptr, col_perm=data.col_perm)
