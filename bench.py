@@ -89,7 +89,9 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
         loss = weight * loss
         mse_log = loss.detach()
         lm = mmd_loss(vloc.permute(0, 2, 1).float(), data.target, data.batch,
-                      data.ptr, data.counts, mmd_sigma, mmd_samples)
+                      data.ptr, data.counts, mmd_sigma, mmd_samples,
+                      sample_idx=getattr(data, "mmd_idx", None),
+                      sample_valid=getattr(data, "mmd_valid", None))
         loss = loss + 0.01 * weight * lm
         (loss / accum).backward()
         return (mse_log,)
@@ -98,10 +100,16 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
 
 
 def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
-               world_size, device, clip=True):
+               world_size, device, clip=True, mmd_cfg=None):
     data = batch.to(device)
     data.counts_global = (comm.global_counts(data.counts)
                           if world_size > 1 else data.counts)
+    if mmd_cfg is not None:
+        # fresh randomness drawn OUTSIDE the captured region
+        from distegnn_amd.runtime.losses import draw_sample_indices
+
+        data.mmd_idx, data.mmd_valid = draw_sample_indices(
+            data.batch, data.ptr, data.counts, mmd_cfg)
     (mse_log,) = graphed(data)
     if (step + 1) % accum == 0:
         if grad_bucket is not None:
@@ -216,17 +224,19 @@ def main():
                           enabled=use_graphs, verbose=True)
 
     clip = args.workload == "largefluid" or world_size > 1
+    num_sample = mmd_samples * vch
     mse = None
     for w in range(args.warmup):
         mse = train_step(graphed, batches[w % len(batches)], optimizer,
-                         grad_bucket, w, accum, world_size, device, clip)
+                         grad_bucket, w, accum, world_size, device, clip,
+                         mmd_cfg=num_sample)
     comm.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for k in range(args.steps):
         mse = train_step(graphed, batches[k % len(batches)], optimizer,
                          grad_bucket, args.warmup + k, accum,
-                         world_size, device, clip)
+                         world_size, device, clip, mmd_cfg=num_sample)
     comm.barrier()
     torch.cuda.synchronize()
     elapsed = torch.tensor(time.perf_counter() - t0, device=device)

@@ -18,8 +18,11 @@ Design:
   (`zero_grad(set_to_none=False)` keeps the addresses stable). The
   optimizer step runs OUTSIDE the graph (it executes every
   accumulation_steps only, on 0.5 MB of parameters).
-* RNG (the MMD sampler's torch.rand/argsort) is captured with the default
-  generator's capture support (philox offsets advance per replay).
+* RNG: random sampling must NOT run inside the captured region — CUDA
+  graph RNG replay of the MMD sampler was observed to corrupt training
+  after ~12 replays on ROCm. Callers draw randomness eagerly per step
+  (losses.draw_sample_indices) and pass it as a static input (mmd_idx /
+  mmd_valid batch fields).
 * Distributed: RCCL collectives inside graphs are gated by
   ``allow_collectives`` (hipGraph capture of RCCL is exercised by the
   multi-GPU bench path; on failure we fall back to eager permanently for
@@ -38,7 +41,8 @@ import torch
 _BATCH_FIELDS = ("x", "pos", "vel", "attr", "target", "edge_index",
                  "edge_attr", "batch", "ptr", "rowptr", "colptr", "col_perm",
                  "counts", "counts_global", "loc_mean", "pool_chunk_begin",
-                 "pool_chunk_end", "pool_seg_chunk_ptr")
+                 "pool_chunk_end", "pool_seg_chunk_ptr", "mmd_idx",
+                 "mmd_valid")
 
 
 class _ShapeEntry:

@@ -37,18 +37,17 @@ def rbf_kernel_sum(x: torch.Tensor, y: torch.Tensor, sigma: float,
     return k.sum()
 
 
-def sample_nodes_per_graph(target: torch.Tensor, batch: torch.Tensor,
-                           ptr: torch.Tensor, counts: torch.Tensor,
-                           num_sample: int):
-    """Device-side per-graph sampling without replacement.
+def draw_sample_indices(batch: torch.Tensor, ptr: torch.Tensor,
+                        counts: torch.Tensor, num_sample: int):
+    """Per-graph without-replacement sample indices (device-side RNG).
 
-    Returns (samples [B, S, 3], valid [B, S] bool). Graphs with fewer than
-    ``num_sample`` nodes contribute all their nodes (reference behavior:
-    randperm[:S] just truncates, train.py:131)."""
-    n = target.size(0)
-    b = ptr.numel() - 1
+    Returns (idx [B*S] long, valid [B, S] bool). Kept SEPARATE from the
+    loss so hipGraph-captured steps can draw fresh randomness eagerly and
+    feed the indices as a static graph input (CUDA-graph RNG replay of the
+    in-loss sampler was observed to corrupt after ~12 replays)."""
+    n = batch.size(0)
     s = num_sample
-    device = target.device
+    device = batch.device
     # Random key within [0,1) + 2*graph_id: argsort groups nodes by graph,
     # randomly permuted inside each graph block.
     keys = torch.rand(n, device=device) + 2.0 * batch.to(torch.float32)
@@ -57,20 +56,40 @@ def sample_nodes_per_graph(target: torch.Tensor, batch: torch.Tensor,
     valid = torch.arange(s, device=device).unsqueeze(0) < counts.unsqueeze(1)
     pos = pos.clamp(max=max(n - 1, 0))
     idx = perm[pos.reshape(-1)]                      # [B*S]
-    samples = target.index_select(0, idx).reshape(b, s, -1)
-    return samples, valid
+    return idx, valid
+
+
+def sample_nodes_per_graph(target: torch.Tensor, batch: torch.Tensor,
+                           ptr: torch.Tensor, counts: torch.Tensor,
+                           num_sample: int, sample_idx=None,
+                           sample_valid=None):
+    """Gather per-graph samples (see draw_sample_indices).
+
+    Returns (samples [B, S, 3], valid [B, S] bool). Graphs with fewer than
+    ``num_sample`` nodes contribute all their nodes (reference behavior:
+    randperm[:S] just truncates, train.py:131)."""
+    b = ptr.numel() - 1
+    if sample_idx is None:
+        sample_idx, sample_valid = draw_sample_indices(batch, ptr, counts,
+                                                       num_sample)
+    samples = target.index_select(0, sample_idx).reshape(b, num_sample, -1)
+    return samples, sample_valid
 
 
 def mmd_loss(virtual_loc_bc3: torch.Tensor, target: torch.Tensor,
              batch: torch.Tensor, ptr: torch.Tensor, counts: torch.Tensor,
-             sigma: float, samples_per_channel: int) -> torch.Tensor:
+             sigma: float, samples_per_channel: int, sample_idx=None,
+             sample_valid=None) -> torch.Tensor:
     """MMD between virtual node positions and the real node distribution.
 
     virtual_loc_bc3: [B, C, 3] (channels-major). Gradient flows into the
-    virtual positions only (targets are data)."""
+    virtual positions only (targets are data). Pass sample_idx/sample_valid
+    (from draw_sample_indices) when running under hipGraph capture."""
     b, c, _ = virtual_loc_bc3.shape
     num_sample = samples_per_channel * c
-    real, valid = sample_nodes_per_graph(target, batch, ptr, counts, num_sample)
+    real, valid = sample_nodes_per_graph(target, batch, ptr, counts,
+                                         num_sample, sample_idx,
+                                         sample_valid)
     l_vv = rbf_kernel_sum(virtual_loc_bc3, virtual_loc_bc3, sigma)
     l_rv = rbf_kernel_sum(real.detach(), virtual_loc_bc3, sigma, mask_x=valid)
     l_vv = l_vv / b / c / c

@@ -35,7 +35,7 @@ from torch import nn
 from ..parallel import comm
 from ..parallel.comm import GradBucket
 from .graphs import GraphedStep
-from .losses import mmd_loss
+from .losses import draw_sample_indices, mmd_loss
 
 try:
     from tqdm import tqdm
@@ -128,7 +128,9 @@ def make_train_step_core(model, model_name, loss_fn, train_config,
             vloc = virtual_node_loc.permute(0, 2, 1).float()
             lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
                           data.counts, train_config.mmd.sigma,
-                          train_config.mmd.samples)
+                          train_config.mmd.samples,
+                          sample_idx=getattr(data, "mmd_idx", None),
+                          sample_valid=getattr(data, "mmd_valid", None))
             loss_loc = loss_loc + train_config.mmd.weight * weight * lm
         (loss_loc / float(train_config.accumulation_steps)).backward()
         return (mse_log,)
@@ -179,6 +181,11 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
             if world_size > 1 else data.counts
 
         if backprop and graphed_step is not None:
+            if _is_fast_model(model_name):
+                # MMD randomness drawn OUTSIDE the hipGraph (static input)
+                ns = train_config.mmd.samples * subgraphs
+                data.mmd_idx, data.mmd_valid = draw_sample_indices(
+                    data.batch, data.ptr, data.counts, ns)
             (mse_log,) = graphed_step(data)
             loss_accum = loss_accum + mse_log * batch_size
             counter = counter + batch_size
