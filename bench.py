@@ -114,11 +114,16 @@ def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
     if (step + 1) % accum == 0:
         if grad_bucket is not None:
             grad_bucket.sync()
-        if clip:  # reference clip rule: FastEGNN + (ws>1 or LargeFluid)
-            torch.nn.utils.clip_grad_norm_(model_params(graphed),
-                                           max_norm=0.3)
-        optimizer.step()
-        optimizer.zero_grad(set_to_none=False)
+
+        def _opt():
+            if clip:  # reference clip rule: FastEGNN + (ws>1 or LargeFluid)
+                torch.nn.utils.clip_grad_norm_(model_params(graphed),
+                                               max_norm=0.3)
+            optimizer.step()
+            optimizer.zero_grad(set_to_none=False)
+
+        # side stream: eager allocs must not alias graph-pool blocks
+        graphed.run_eager(_opt)
     return mse_log
 
 

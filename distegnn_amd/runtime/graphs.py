@@ -18,11 +18,18 @@ Design:
   (`zero_grad(set_to_none=False)` keeps the addresses stable). The
   optimizer step runs OUTSIDE the graph (it executes every
   accumulation_steps only, on 0.5 MB of parameters).
-* RNG: random sampling must NOT run inside the captured region — CUDA
-  graph RNG replay of the MMD sampler was observed to corrupt training
-  after ~12 replays on ROCm. Callers draw randomness eagerly per step
-  (losses.draw_sample_indices) and pass it as a static input (mmd_idx /
-  mmd_valid batch fields).
+* RNG: random sampling runs OUTSIDE the captured region — callers draw
+  per-step randomness eagerly (losses.draw_sample_indices) and pass it as
+  a static input (mmd_idx / mmd_valid batch fields). In-graph RNG would
+  replay a frozen philox offset, repeating the same "random" sample every
+  step.
+* Eager work BETWEEN replays (optimizer step, grad clip) must go through
+  ``run_eager``: on this ROCm stack the allocator was observed to hand
+  default-stream blocks to eager ops while a captured graph's private
+  pool still used them (Adam temporaries aliased graph blocks after ~11
+  replays — loss scaled wrong, params eventually NaN). ``run_eager``
+  runs the callable on a dedicated side stream whose per-stream blocks
+  never overlap the graph pool.
 * Distributed: RCCL collectives inside graphs are gated by
   ``allow_collectives`` (hipGraph capture of RCCL is exercised by the
   multi-GPU bench path; on failure we fall back to eager permanently for
@@ -50,7 +57,8 @@ class _ShapeEntry:
         self.seen = 0
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.static = None       # dict of static input buffers
-        self.outputs = None      # (loss, mse) static outputs
+        self.outputs = None      # eager-owned output buffers
+        self.out_meta = None     # [(shape, dtype), ...] from last warmup
         self.disabled = False
 
 
@@ -69,6 +77,31 @@ class GraphedStep:
         self.enabled = enabled and torch.cuda.is_available()
         self.verbose = verbose
         self.entries: Dict[Tuple, _ShapeEntry] = {}
+        self._pool = None  # shared mempool across shape keys (see capture)
+        self._side = None  # side stream for eager work between replays
+
+    def run_eager(self, fn):
+        """Run eager work (optimizer step, clip, zero_grad) between replays.
+
+        MUST be used for any eager op that allocates device memory between
+        replays of a captured graph. On this ROCm stack the caching
+        allocator hands default-stream blocks to eager ops even though a
+        captured graph's private pool still references them: Adam's
+        foreach temporaries started aliasing graph blocks after ~11
+        replays, silently scaling the loss (and eventually NaN-ing
+        parameters). Running the eager work on a dedicated side stream
+        keeps its allocations in per-stream blocks that never overlap the
+        graph pool (verified: 30 replays bit-exact vs eager,
+        tools/graph_bisect3.py adam_side)."""
+        if not self.enabled:
+            return fn()
+        if self._side is None:
+            self._side = torch.cuda.Stream()
+        self._side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._side):
+            out = fn()
+        torch.cuda.current_stream().wait_stream(self._side)
+        return out
 
     @staticmethod
     def _key(batch) -> Tuple:
@@ -98,8 +131,11 @@ class GraphedStep:
             e.graph.replay()
             return e.outputs
         e.seen += 1
-        if e.seen <= self.warmup:
-            return self.step_fn(batch)
+        if e.seen <= self.warmup or e.out_meta is None:
+            outs = self.step_fn(batch)
+            e.out_meta = [(tuple(o.shape), o.dtype, o.device)
+                          for o in outs]
+            return outs
         # capture attempt
         try:
             static = self._snapshot(batch)
@@ -118,14 +154,24 @@ class GraphedStep:
             for p in self.params:
                 if p.grad is None:
                     p.grad = torch.zeros_like(p)
+            # Output buffers are allocated EAGERLY (normal allocator pool)
+            # before capture and the captured region copies into them —
+            # exactly like the pre-materialized .grad buffers. Returning
+            # tensors that live in the graph's private mempool is unsafe on
+            # this ROCm stack: post-replay eager work (Adam temporaries)
+            # was observed to scribble private-pool blocks, corrupting the
+            # logged loss while training itself stayed correct.
+            out_bufs = tuple(torch.empty(shape, dtype=dtype, device=dev)
+                             for (shape, dtype, dev) in e.out_meta)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
-            # warm side stream, then capture
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, pool=self._pool):
                 outputs = self.step_fn(proxy)
+                for buf, o in zip(out_bufs, outputs):
+                    buf.copy_(o)
             e.graph = g
             e.static = static
-            e.outputs = outputs
+            e.outputs = out_bufs
             if self.verbose:
                 print(f"[graphs] captured step for shape {key}")
             # the capture itself already executed once? No: capture does

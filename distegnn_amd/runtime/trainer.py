@@ -222,13 +222,23 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
             if (step + 1) % train_config.accumulation_steps == 0:
                 if grad_bucket is not None:
                     grad_bucket.sync()              # SUM == ref avg * ws
-                if ((world_size > 1 or dataset_name == "LargeFluid")
-                        and model_name == "FastEGNN"):
-                    nn.utils.clip_grad_norm_(model.parameters(), max_norm=0.3)
-                optimizer.step()
-                if scheduler is not None:
-                    scheduler.step()
-                optimizer.zero_grad(set_to_none=False)
+
+                def _opt_region():
+                    if ((world_size > 1 or dataset_name == "LargeFluid")
+                            and model_name == "FastEGNN"):
+                        nn.utils.clip_grad_norm_(model.parameters(),
+                                                 max_norm=0.3)
+                    optimizer.step()
+                    if scheduler is not None:
+                        scheduler.step()
+                    optimizer.zero_grad(set_to_none=False)
+
+                if graphed_step is not None:
+                    # side stream: eager optimizer allocs must not alias
+                    # captured-graph pool blocks (see GraphedStep.run_eager)
+                    graphed_step.run_eager(_opt_region)
+                else:
+                    _opt_region()
         if step_timer is not None:
             step_timer.stop()
 
