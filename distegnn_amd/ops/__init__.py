@@ -59,6 +59,17 @@ def _require_ext(opname: str):
     return ext
 
 
+def _fast_gather(x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """dst[i] = x[idx[i]] via the vectorized HIP gather (falls back to
+    index_select off-GPU or for rows not 4-byte aligned)."""
+    ext = _load_extension()
+    if ext is not None and x.is_cuda:
+        row_bytes = (x.numel() // max(x.size(0), 1)) * x.element_size()
+        if row_bytes % 4 == 0 and x.dim() >= 1:
+            return ext.gather_rows_fast(x, idx)
+    return x.index_select(0, idx)
+
+
 class _SegmentReduceFn(torch.autograd.Function):
     """CSR segmented sum/mean over row-sorted edge data (HIP forward)."""
 
@@ -81,7 +92,7 @@ class _SegmentReduceFn(torch.autograd.Function):
         if ctx.mean:
             deg = (rowptr[1:] - rowptr[:-1]).clamp(min=1).to(gout.dtype)
             gout = gout / deg.unsqueeze(-1)
-        gdata = gout.index_select(0, row)
+        gdata = _fast_gather(gout, row)
         return gdata, None, None, None, None
 
 
@@ -130,7 +141,7 @@ class _GraphPoolFn(torch.autograd.Function):
         if ctx.mean:
             cnt = (ptr[1:] - ptr[:-1]).clamp(min=1).to(gout.dtype)
             gout = gout / cnt.view(-1, *([1] * (gout.dim() - 1)))
-        gx = gout.index_select(0, batch)
+        gx = _fast_gather(gout, batch)
         return gx, None, None, None, None, None
 
 
@@ -179,6 +190,8 @@ class _GatherRowsFn(torch.autograd.Function):
                               seg_chunk_ptr if seg_chunk_ptr is not None
                               else idx.new_empty(0))
         ctx.x_rows = x.size(0)
+        if x.is_cuda:
+            return _fast_gather(x, idx)
         return x.index_select(0, idx)
 
     @staticmethod
@@ -186,12 +199,17 @@ class _GatherRowsFn(torch.autograd.Function):
         idx, segptr, perm, cb, ce, scp = ctx.saved_tensors
         ext = _require_ext("gather_rows.backward")
         g = gout.contiguous()
-        if perm.numel():
-            g = g.index_select(0, perm)
         if ext is None:
+            if perm.numel():
+                g = g.index_select(0, perm)
             gx = reference.segment_sum(g, idx, ctx.x_rows)
         elif cb.numel():
+            if perm.numel():
+                g = _fast_gather(g, perm)
             gx = ext.segment_reduce_chunked(g, segptr, cb, ce, scp, False)
+        elif perm.numel():
+            # fold the col-sort permutation into the reduction itself
+            gx = ext.segment_reduce_csr_perm(g, segptr, perm, False)
         else:
             gx = ext.segment_reduce_csr(g, segptr, False)
         return gx, None, None, None, None, None, None
@@ -278,11 +296,11 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
                 w1.bfloat16(), b1, w2.bfloat16(), b2, w3.bfloat16(), b3,
                 w3v, bool(ctx.normalize), float(ctx.eps))
             gh = (ext.segment_reduce_csr(dhr, rowptr, False)
-                  + ext.segment_reduce_csr(dhc.index_select(0, col_perm),
-                                           colptr, False))
+                  + ext.segment_reduce_csr_perm(dhc, colptr, col_perm,
+                                                False))
             gc = (ext.segment_reduce_csr(dcd, rowptr, False)
-                  - ext.segment_reduce_csr(dcd.index_select(0, col_perm),
-                                           colptr, False))
+                  - ext.segment_reduce_csr_perm(dcd, colptr, col_perm,
+                                                False))
             from .linear import chunked_wgrad
 
             k_in = w1.size(1)

@@ -3,6 +3,10 @@
 
 torch::Tensor segment_reduce_csr(torch::Tensor data, torch::Tensor rowptr,
                                  bool mean);
+torch::Tensor segment_reduce_csr_perm(torch::Tensor data,
+                                      torch::Tensor rowptr,
+                                      torch::Tensor perm, bool mean);
+torch::Tensor gather_rows_fast(torch::Tensor data, torch::Tensor idx);
 torch::Tensor segment_reduce_chunked(torch::Tensor data, torch::Tensor rowptr,
                                      torch::Tensor chunk_begin,
                                      torch::Tensor chunk_end,
@@ -46,6 +50,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_reduce_csr", &segment_reduce_csr,
         "deterministic CSR segmented sum/mean", py::arg("data"),
         py::arg("rowptr"), py::arg("mean"));
+  m.def("segment_reduce_csr_perm", &segment_reduce_csr_perm,
+        "CSR segmented sum/mean over permuted rows (data[perm[k]])",
+        py::arg("data"), py::arg("rowptr"), py::arg("perm"), py::arg("mean"));
+  m.def("gather_rows_fast", &gather_rows_fast,
+        "vectorized dst[i] = src[idx[i]] row gather", py::arg("data"),
+        py::arg("idx"));
   m.def("segment_reduce_chunked", &segment_reduce_chunked,
         "two-stage deterministic segmented reduce for huge segments",
         py::arg("data"), py::arg("rowptr"), py::arg("chunk_begin"),

@@ -26,9 +26,14 @@
 
 namespace {
 
+// All variants take an optional row permutation: row k of the CSR order
+// reads data[perm[k]] (perm == nullptr -> identity). This folds the
+// "sort columns into col-CSR order" gather into the reduction itself,
+// removing a materialized [M, F] permuted copy per backward.
 template <typename T>
 __global__ void seg_reduce_elem(const T* __restrict__ data,
                                 const long* __restrict__ rowptr,
+                                const long* __restrict__ perm,
                                 T* __restrict__ out, long n, int f,
                                 bool mean) {
   long total = n * f;
@@ -38,7 +43,10 @@ __global__ void seg_reduce_elem(const T* __restrict__ data,
     int j = (int)(o - seg * f);
     long s = rowptr[seg], e = rowptr[seg + 1];
     float acc = 0.f;
-    for (long k = s; k < e; ++k) acc += to_f32<T>(data[k * f + j]);
+    for (long k = s; k < e; ++k) {
+      long kr = perm ? perm[k] : k;
+      acc += to_f32<T>(data[kr * f + j]);
+    }
     if (mean && e > s) acc /= (float)(e - s);
     out[o] = from_f32<T>(acc);
   }
@@ -47,6 +55,7 @@ __global__ void seg_reduce_elem(const T* __restrict__ data,
 template <typename T>
 __global__ void seg_reduce_wave(const T* __restrict__ data,
                                 const long* __restrict__ rowptr,
+                                const long* __restrict__ perm,
                                 T* __restrict__ out, long n, int f,
                                 bool mean) {
   int lane = threadIdx.x & (WAVE - 1);
@@ -57,7 +66,10 @@ __global__ void seg_reduce_wave(const T* __restrict__ data,
     float inv = (mean && e > s) ? 1.f / (float)(e - s) : 1.f;
     for (int j = lane; j < f; j += WAVE) {
       float acc = 0.f;
-      for (long k = s; k < e; ++k) acc += to_f32<T>(data[k * f + j]);
+      for (long k = s; k < e; ++k) {
+        long kr = perm ? perm[k] : k;
+        acc += to_f32<T>(data[kr * f + j]);
+      }
       out[seg * f + j] = from_f32<T>(acc * inv);
     }
   }
@@ -70,6 +82,7 @@ __global__ void seg_reduce_wave(const T* __restrict__ data,
 template <typename T>
 __global__ void seg_reduce_wave4(const T* __restrict__ data,
                                  const long* __restrict__ rowptr,
+                                 const long* __restrict__ perm,
                                  T* __restrict__ out, long n, int f,
                                  bool mean) {
   int lane = threadIdx.x & (WAVE - 1);
@@ -84,7 +97,8 @@ __global__ void seg_reduce_wave4(const T* __restrict__ data,
     for (int fq = fl; fq < fquads; fq += 16) {
       float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
       for (long k = s + rl; k < e; k += 4) {
-        const T* p = data + k * f + fq * 4;
+        long kr = perm ? perm[k] : k;
+        const T* p = data + kr * f + fq * 4;
         a0 += to_f32<T>(p[0]);
         a1 += to_f32<T>(p[1]);
         a2 += to_f32<T>(p[2]);
@@ -175,14 +189,21 @@ struct hip_type<at::BFloat16> {
 
 }  // namespace
 
-torch::Tensor segment_reduce_csr(torch::Tensor data, torch::Tensor rowptr,
-                                 bool mean) {
+torch::Tensor segment_reduce_csr_perm(torch::Tensor data,
+                                      torch::Tensor rowptr,
+                                      torch::Tensor perm, bool mean) {
   TORCH_CHECK(data.is_cuda() && rowptr.is_cuda(), "expected CUDA tensors");
   TORCH_CHECK(rowptr.scalar_type() == torch::kLong, "rowptr must be int64");
   auto d = data.contiguous();
   auto rp = rowptr.contiguous();
+  const long* pp = nullptr;
+  torch::Tensor pc;
+  if (perm.defined() && perm.numel() > 0) {
+    TORCH_CHECK(perm.scalar_type() == torch::kLong, "perm must be int64");
+    pc = perm.contiguous();
+    pp = pc.data_ptr<long>();
+  }
   long n = rp.numel() - 1;
-  long m = d.size(0);
   long f = 1;
   for (int i = 1; i < d.dim(); ++i) f *= d.size(i);
   std::vector<int64_t> oshape(d.sizes().begin(), d.sizes().end());
@@ -200,16 +221,82 @@ torch::Tensor segment_reduce_csr(torch::Tensor data, torch::Tensor rowptr,
         if (f >= 16 && (f & 3) == 0) {
           int threads = 256;
           seg_reduce_wave4<T><<<num_blocks(n * WAVE, threads), threads, 0,
-                                stream>>>(dp, rpp, op, n, (int)f, mean);
+                                stream>>>(dp, rpp, pp, op, n, (int)f, mean);
         } else if (f >= 16) {
           int threads = 256;
           seg_reduce_wave<T><<<num_blocks(n * WAVE, threads), threads, 0,
-                               stream>>>(dp, rpp, op, n, (int)f, mean);
+                               stream>>>(dp, rpp, pp, op, n, (int)f, mean);
         } else {
           seg_reduce_elem<T><<<num_blocks(n * f, 256), 256, 0, stream>>>(
-              dp, rpp, op, n, (int)f, mean);
+              dp, rpp, pp, op, n, (int)f, mean);
         }
       });
+  return out;
+}
+
+torch::Tensor segment_reduce_csr(torch::Tensor data, torch::Tensor rowptr,
+                                 bool mean) {
+  return segment_reduce_csr_perm(data, rowptr, torch::Tensor(), mean);
+}
+
+namespace {
+
+// Vectorized row gather: dst[i] = src[idx[i]] with 16 B / 8 B word copies.
+// Coalesced writes, b128 reads; replaces aten's vectorized_gather_kernel
+// (216 us/call for a [1.65M, 64] bf16 gather vs ~60 us roofline).
+template <typename W>
+__global__ void gather_rows_words(const W* __restrict__ src,
+                                  const long* __restrict__ idx,
+                                  W* __restrict__ dst, long n_out,
+                                  int words) {
+  long total = n_out * (long)words;
+  for (long o = blockIdx.x * (long)blockDim.x + threadIdx.x; o < total;
+       o += (long)gridDim.x * blockDim.x) {
+    long i = o / words;
+    int w = (int)(o - i * (long)words);
+    dst[o] = src[idx[i] * (long)words + w];
+  }
+}
+
+}  // namespace
+
+torch::Tensor gather_rows_fast(torch::Tensor data, torch::Tensor idx) {
+  TORCH_CHECK(data.is_cuda() && idx.is_cuda(), "expected CUDA tensors");
+  TORCH_CHECK(idx.scalar_type() == torch::kLong, "idx must be int64");
+  auto d = data.contiguous();
+  auto ix = idx.contiguous();
+  long n_out = ix.numel();
+  long row_bytes = (d.numel() / std::max<long>(d.size(0), 1)) *
+                   d.element_size();
+  std::vector<int64_t> oshape(d.sizes().begin(), d.sizes().end());
+  oshape[0] = n_out;
+  auto out = torch::empty(oshape, d.options());
+  if (n_out == 0 || row_bytes == 0) return out;
+  auto stream = at::hip::getCurrentHIPStream();
+  const long* ip = ix.data_ptr<long>();
+  if (row_bytes % 16 == 0) {
+    int words = (int)(row_bytes / 16);
+    using W = ulonglong2;
+    gather_rows_words<W><<<num_blocks(n_out * words, 256), 256, 0, stream>>>(
+        reinterpret_cast<const W*>(d.data_ptr()), ip,
+        reinterpret_cast<W*>(out.data_ptr()), n_out, words);
+  } else if (row_bytes % 8 == 0) {
+    int words = (int)(row_bytes / 8);
+    gather_rows_words<unsigned long long>
+        <<<num_blocks(n_out * words, 256), 256, 0, stream>>>(
+            reinterpret_cast<const unsigned long long*>(d.data_ptr()), ip,
+            reinterpret_cast<unsigned long long*>(out.data_ptr()), n_out,
+            (int)words);
+  } else if (row_bytes % 4 == 0) {
+    int words = (int)(row_bytes / 4);
+    gather_rows_words<unsigned int>
+        <<<num_blocks(n_out * words, 256), 256, 0, stream>>>(
+            reinterpret_cast<const unsigned int*>(d.data_ptr()), ip,
+            reinterpret_cast<unsigned int*>(out.data_ptr()), n_out,
+            (int)words);
+  } else {
+    TORCH_CHECK(false, "gather_rows_fast: row bytes must be 4-aligned");
+  }
   return out;
 }
 
