@@ -291,7 +291,7 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             dtrans_n = (dagg_trans
                         / deg.unsqueeze(-1).to(dagg_trans.dtype)).contiguous()
             (ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd,
-             gw3v) = ext.fused_edge_backward(
+             gw3v, gb) = ext.fused_edge_backward(
                 h, coord, eattr, row, col, dmsg_n, dtrans_n,
                 w1.bfloat16(), b1, w2.bfloat16(), b2, w3.bfloat16(), b3,
                 w3v, bool(ctx.normalize), float(ctx.eps))
@@ -307,9 +307,9 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             gw1 = chunked_wgrad(dz1, ein)[:, :k_in].float()
             gw2 = chunked_wgrad(dz2, t1).float()
             gw3 = chunked_wgrad(dz3, msg).float()
-            gb1 = dz1.sum(0).float()
-            gb2 = dz2.sum(0).float()
-            gb3 = dz3.sum(0).float()
+            # bias grads accumulated in-kernel (block LDS + atomics):
+            # avoids three aten column-sum re-reads of [M, 64]
+            gb1, gb2, gb3 = gb[:64], gb[64:128], gb[128:]
             return (gh.to(h.dtype), gc, None, None, None, None, None, None,
                     gw1, gb1, gw2, gb2, gw3, gb3, gw3v, None, None)
         # fallback: recompute the eager composition under autograd
@@ -423,7 +423,7 @@ class _FusedVirtualBlockFn(torch.autograd.Function):
         n, c = ctx.shape_nc
         rows = n * c
         (dz1, dz2, dzxv, dzX, dh_row, dvf_row, dgram_row, dvd,
-         dp2) = ext.fused_virtual_backward(
+         dp2, gb) = ext.fused_virtual_backward(
             coord, vcoord.float(), batch,
             dvmsg.reshape(rows, -1).to(torch.bfloat16).contiguous(),
             dtv.reshape(rows, 3).float().contiguous(),
@@ -438,14 +438,11 @@ class _FusedVirtualBlockFn(torch.autograd.Function):
         gw2 = chunked_wgrad(dz2, t1).float()
         gwxv = chunked_wgrad(dzxv, vmsg).float()
         gwX = chunked_wgrad(dzX, vmsg).float()
-        gb1 = dz1.sum(0, dtype=torch.float32)
-        gb2 = dz2.sum(0, dtype=torch.float32)
-        gbxv = dzxv.sum(0, dtype=torch.float32)
-        gbX = dzX.sum(0, dtype=torch.float32)
-        sxv = torch.nn.functional.silu(zxv)
-        s_x = torch.nn.functional.silu(zX)
-        gwxvv = (sxv.float() * dp2[:, :1]).sum(0)
-        gwXv = (s_x.float() * dp2[:, 1:]).sum(0)
+        # bias + head-weight grads accumulated in-kernel (LDS + atomics):
+        # avoids four [R,64] column-sum re-reads and two full-tensor silus
+        gb1, gb2 = gb[:64], gb[64:128]
+        gbxv, gbX = gb[128:192], gb[192:256]
+        gwxvv, gwXv = gb[256:320], gb[320:384]
 
         gh = dh_row.view(n, c, -1).sum(1, dtype=torch.float32).to(h.dtype)
         gcoord = -dvd.view(n, c, 3).sum(1)

@@ -58,6 +58,7 @@ struct Smem {
   int scal;     // [TILE][4] f32 (p, dp/dr2, -, -)
   int bias;     // [4*H] f32
   int wpart;    // [H] f32
+  int gbacc;    // [3*H] f32 (block-local gb1/gb2/gb3 column sums)
   int total;
 };
 
@@ -72,6 +73,7 @@ __host__ __device__ constexpr Smem smem_layout() {
   L.scal = o; o += TILE * 4 * 4;
   L.bias = o; o += 4 * H * 4;
   L.wpart = o; o += H * 4;
+  L.gbacc = o; o += 3 * H * 4;
   L.total = o;
   return L;
 }
@@ -134,8 +136,9 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
     bf16* __restrict__ msg_out, bf16* __restrict__ dz1_out,
     bf16* __restrict__ dz2_out, bf16* __restrict__ dz3_out,
     bf16* __restrict__ dhr_out, bf16* __restrict__ dhc_out,
-    float* __restrict__ dcd_out, float* __restrict__ dw3v_out, long m,
-    int normalize, float eps) {
+    float* __restrict__ dcd_out, float* __restrict__ dw3v_out,
+    float* __restrict__ gb_out,  // [3*H]: gb1 | gb2 | gb3 column sums
+    long m, int normalize, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr Smem L = smem_layout();
   const int tid = threadIdx.x;
@@ -144,12 +147,16 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
 
   float* biases = reinterpret_cast<float*>(smem + L.bias);
   float* wpart = reinterpret_cast<float*>(smem + L.wpart);
+  float* gbacc = reinterpret_cast<float*>(smem + L.gbacc);
   for (int i = tid; i < H; i += THREADS) {
     biases[i] = b1[i];
     biases[H + i] = b2[i];
     biases[2 * H + i] = b3[i];
     biases[3 * H + i] = w3v[i];
     wpart[i] = 0.f;
+    gbacc[i] = 0.f;
+    gbacc[H + i] = 0.f;
+    gbacc[2 * H + i] = 0.f;
   }
 
   for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
@@ -300,7 +307,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       // thread covers column c over 16 edges: dw3v partial + dz3 in place
       int c = tid & 63;
       int estart = (tid >> 6) * 16;
-      float acc_w = 0.f;
+      float acc_w = 0.f, acc_b3 = 0.f;
       float wv = biases[3 * H + c];
       __bf16* z3 = reinterpret_cast<__bf16*>(smem + L.z3);
       const float* sc = reinterpret_cast<const float*>(smem + L.scal);
@@ -308,9 +315,12 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
         float z = (float)z3[e * H_STRIDE + c];
         float dp = sc[e * 4 + 1];
         acc_w += dp * silu_(z);
-        z3[e * H_STRIDE + c] = (__bf16)(dp * wv * dsilu_(z));
+        float d3 = dp * wv * dsilu_(z);
+        acc_b3 += d3;
+        z3[e * H_STRIDE + c] = (__bf16)d3;
       }
       atomicAdd(&wpart[c], acc_w);
+      atomicAdd(&gbacc[2 * H + c], acc_b3);
     }
     __syncthreads();
     for (int idx = tid; idx < TILE * 8; idx += THREADS) {
@@ -350,6 +360,15 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
           lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
     }
+    {
+      int c = tid & 63;
+      int estart = (tid >> 6) * 16;
+      const __bf16* z2 = reinterpret_cast<const __bf16*>(smem + L.z2);
+      float acc_b = 0.f;
+      for (int e = estart; e < estart + 16; ++e)
+        acc_b += (float)z2[e * H_STRIDE + c];
+      atomicAdd(&gbacc[H + c], acc_b);
+    }
 
     // ---- dz1 = (dz2 @ W2) silu'(z1), overwrite z1 ----
     {
@@ -376,6 +395,15 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
           lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+    }
+    {
+      int c = tid & 63;
+      int estart = (tid >> 6) * 16;
+      const __bf16* z1 = reinterpret_cast<const __bf16*>(smem + L.z1);
+      float acc_b = 0.f;
+      for (int e = estart; e < estart + 16; ++e)
+        acc_b += (float)z1[e * H_STRIDE + c];
+      atomicAdd(&gbacc[c], acc_b);
     }
 
     // ---- dein = dz1 @ W1 in 3 register passes of 3 n-tiles ----
@@ -440,6 +468,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
   }
   __syncthreads();
   for (int c = tid; c < H; c += THREADS) atomicAdd(&dw3v_out[c], wpart[c]);
+  for (int c = tid; c < 3 * H; c += THREADS) atomicAdd(&gb_out[c], gbacc[c]);
 }
 
 }  // namespace
@@ -463,7 +492,8 @@ std::vector<torch::Tensor> fused_edge_backward(
   auto dhc = torch::empty({m, (long)H}, bopt);
   auto dcd = torch::empty({m, 3}, fopt);
   auto dw3v = torch::zeros({(long)H}, fopt);
-  if (m == 0) return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v};
+  auto gb = torch::zeros({3 * (long)H}, fopt);
+  if (m == 0) return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
   auto stream = at::hip::getCurrentHIPStream();
   constexpr Smem L = smem_layout();
   long tiles = (m + TILE - 1) / TILE;
@@ -505,6 +535,7 @@ std::vector<torch::Tensor> fused_edge_backward(
       reinterpret_cast<bf16*>(dz3.data_ptr()),
       reinterpret_cast<bf16*>(dhr.data_ptr()),
       reinterpret_cast<bf16*>(dhc.data_ptr()), dcd.data_ptr<float>(),
-      dw3v.data_ptr<float>(), m, normalize ? 1 : 0, (float)eps);
-  return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v};
+      dw3v.data_ptr<float>(), gb.data_ptr<float>(), m, normalize ? 1 : 0,
+      (float)eps);
+  return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
 }

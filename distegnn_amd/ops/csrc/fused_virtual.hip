@@ -93,6 +93,7 @@ struct VSmem {
   int diff;     // [TILE][4] f32 (vdiff xyz, vrad)
   int scal;     // [TILE][4] f32 (pxv, pX / dpxv, dr)
   int bias;     // [6*H] f32 (b1, b2, bxv, bX, wxv, wX)
+  int gbacc;    // [6*H] f32 (bwd: gb1, gb2, gbxv, gbX, gwxvv, gwXv)
   int total;
 };
 
@@ -107,6 +108,7 @@ __host__ __device__ constexpr VSmem vsmem_layout() {
   L.diff = o; o += TILE * 4 * 4;
   L.scal = o; o += TILE * 4 * 4;
   L.bias = o; o += 6 * H * 4;
+  L.gbacc = o; o += 6 * H * 4;
   L.total = o;
   return L;
 }
@@ -337,6 +339,7 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     float* __restrict__ dgram_out,     // [R,CMAX]
     float* __restrict__ dvd_out,       // [R,3]
     float* __restrict__ dp2_out,       // [R,2]
+    float* __restrict__ gb_out,  // [6H]: gb1|gb2|gbxv|gbX|gwxvv|gwXv
     long n_rows, int cdim, int k_in) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr VSmem L = vsmem_layout();
@@ -345,9 +348,11 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
   const int wave = tid >> 6;
 
   float* biases = reinterpret_cast<float*>(smem + L.bias);
+  float* gbacc = reinterpret_cast<float*>(smem + L.gbacc);
   for (int i = tid; i < H; i += THREADS) {
     biases[4 * H + i] = wxvv[i];
     biases[5 * H + i] = wXv[i];
+    for (int kacc = 0; kacc < 6; ++kacc) gbacc[kacc * H + i] = 0.f;
   }
 
   for (long tile = blockIdx.x; tile * TILE < n_rows; tile += gridDim.x) {
@@ -390,14 +395,28 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       __bf16* zc = reinterpret_cast<__bf16*>(smem + L.zc);
       __bf16* zd = reinterpret_cast<__bf16*>(smem + L.zd);
       const float* sc = reinterpret_cast<const float*>(smem + L.scal);
+      float abxv = 0.f, abX = 0.f, awxvv = 0.f, awXv = 0.f;
       for (int e = estart; e < estart + 16; ++e) {
         long r = r0 + e;
         bool ok = e < nrow;
         float zx = ok ? (float)((const __bf16*)zxv_in)[r * H + cc] : 0.f;
         float zX_ = ok ? (float)((const __bf16*)zX_in)[r * H + cc] : 0.f;
-        zc[e * H_STRIDE + cc] = (__bf16)(sc[e * 4] * wv0 * dsilu_(zx));
-        zd[e * H_STRIDE + cc] = (__bf16)(sc[e * 4 + 1] * wv1 * dsilu_(zX_));
+        float dpxv = sc[e * 4], dpX = sc[e * 4 + 1];
+        float dxv = dpxv * wv0 * dsilu_(zx);
+        float dX = dpX * wv1 * dsilu_(zX_);
+        zc[e * H_STRIDE + cc] = (__bf16)dxv;
+        zd[e * H_STRIDE + cc] = (__bf16)dX;
+        abxv += dxv;
+        abX += dX;
+        if (ok) {
+          awxvv += silu_(zx) * dpxv;
+          awXv += silu_(zX_) * dpX;
+        }
       }
+      atomicAdd(&gbacc[2 * H + cc], abxv);
+      atomicAdd(&gbacc[3 * H + cc], abX);
+      atomicAdd(&gbacc[4 * H + cc], awxvv);
+      atomicAdd(&gbacc[5 * H + cc], awXv);
     }
     __syncthreads();
     TILE_TO_GLOBAL(L.zc, dzxv_out, false);
@@ -430,6 +449,15 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     }
     __syncthreads();
     TILE_TO_GLOBAL(L.zb, dz2_out, false);
+    {
+      int cc = tid & 63;
+      int estart = (tid >> 6) * 16;
+      const __bf16* zb = reinterpret_cast<const __bf16*>(smem + L.zb);
+      float ab = 0.f;
+      for (int e = estart; e < estart + 16; ++e)
+        ab += (float)zb[e * H_STRIDE + cc];
+      atomicAdd(&gbacc[H + cc], ab);
+    }
 
     // dz1 = (dz2 @ W2) o silu'(z1)
     {
@@ -452,6 +480,15 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     }
     __syncthreads();
     TILE_TO_GLOBAL(L.za, dz1_out, false);
+    {
+      int cc = tid & 63;
+      int estart = (tid >> 6) * 16;
+      const __bf16* za = reinterpret_cast<const __bf16*>(smem + L.za);
+      float ab = 0.f;
+      for (int e = estart; e < estart + 16; ++e)
+        ab += (float)za[e * H_STRIDE + cc];
+      atomicAdd(&gbacc[cc], ab);
+    }
 
     // dvin = dz1 @ W1 (3 register passes x 3 n-tiles), into in_tile
 #pragma unroll
@@ -521,6 +558,9 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
                            + dptr[2] * inv;
     }
   }
+  __syncthreads();
+  for (int c = tid; c < 6 * H; c += THREADS)
+    atomicAdd(&gb_out[c], gbacc[c]);
 }
 
 }  // namespace
@@ -619,8 +659,9 @@ std::vector<torch::Tensor> fused_virtual_backward(
   auto dgram = torch::empty({rows, (long)CMAX}, fopt);
   auto dvd = torch::empty({rows, 3}, fopt);
   auto dp2 = torch::empty({rows, 2}, fopt);
+  auto gb = torch::zeros({6 * (long)H}, fopt);
   if (rows == 0)
-    return {dz1, dz2, dzxv, dzX, dh, dvf, dgram, dvd, dp2};
+    return {dz1, dz2, dzxv, dzX, dh, dvf, dgram, dvd, dp2, gb};
   auto stream = at::hip::getCurrentHIPStream();
   constexpr VSmem L = vsmem_layout();
   long tiles = (rows + TILE - 1) / TILE;
@@ -656,6 +697,7 @@ std::vector<torch::Tensor> fused_virtual_backward(
       reinterpret_cast<bf16*>(dzX.data_ptr()),
       reinterpret_cast<bf16*>(dh.data_ptr()),
       reinterpret_cast<bf16*>(dvf.data_ptr()), dgram.data_ptr<float>(),
-      dvd.data_ptr<float>(), dp2.data_ptr<float>(), rows, cdim, k_in);
-  return {dz1, dz2, dzxv, dzX, dh, dvf, dgram, dvd, dp2};
+      dvd.data_ptr<float>(), dp2.data_ptr<float>(), gb.data_ptr<float>(),
+      rows, cdim, k_in);
+  return {dz1, dz2, dzxv, dzX, dh, dvf, dgram, dvd, dp2, gb};
 }
