@@ -142,6 +142,70 @@ __global__ void seg_reduce_chunk(const T* __restrict__ data,
   }
 }
 
+// small-f path (f <= 16): one WAVE per chunk, FL feature-lanes x RL
+// row-sublanes, fixed-shape shfl_xor tree over the row sublanes
+// (deterministic). Keeps the whole GPU busy for f like 3/15 where a
+// thread-per-output mapping yields only a few thousand threads.
+template <typename T, int FL>
+__global__ void seg_reduce_chunk_small(const T* __restrict__ data,
+                                       const long* __restrict__ chunk_begin,
+                                       const long* __restrict__ chunk_end,
+                                       float* __restrict__ partial,
+                                       long nchunks, int f) {
+  constexpr int RL = WAVE / FL;
+  int lane = threadIdx.x & (WAVE - 1);
+  int rl = lane / FL;
+  int fl = lane % FL;
+  long wave = (blockIdx.x * (long)blockDim.x + threadIdx.x) / WAVE;
+  long nwaves = ((long)gridDim.x * blockDim.x) / WAVE;
+  for (long c = wave; c < nchunks; c += nwaves) {
+    long s = chunk_begin[c], e = chunk_end[c];
+    for (int j = fl; j < f; j += FL) {
+      float acc = 0.f;
+      for (long k = s + rl; k < e; k += RL) acc += to_f32<T>(data[k * f + j]);
+#pragma unroll
+      for (int off = FL; off < WAVE; off <<= 1) acc += __shfl_xor(acc, off);
+      if (rl == 0) partial[c * f + j] = acc;
+    }
+  }
+}
+
+// f % 4 == 0 fast path: thread covers 4 contiguous features (b64 loads for
+// bf16) with two rows unrolled for MLP; same partial layout as above.
+template <typename T>
+__global__ void seg_reduce_chunk4(const T* __restrict__ data,
+                                  const long* __restrict__ chunk_begin,
+                                  const long* __restrict__ chunk_end,
+                                  float* __restrict__ partial, long nchunks,
+                                  int f) {
+  int fquads = f >> 2;
+  long ftiles = (fquads + blockDim.x - 1) / blockDim.x;
+  for (long b = blockIdx.x; b < nchunks * ftiles; b += gridDim.x) {
+    long c = b / ftiles;
+    int fq = (int)(b - c * ftiles) * blockDim.x + threadIdx.x;
+    if (fq >= fquads) continue;
+    long s = chunk_begin[c], e = chunk_end[c];
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    float b0 = 0.f, b1 = 0.f, b2 = 0.f, b3 = 0.f;
+    long k = s;
+    for (; k + 1 < e; k += 2) {
+      const T* p = data + k * f + fq * 4;
+      const T* q = p + f;
+      a0 += to_f32<T>(p[0]); a1 += to_f32<T>(p[1]);
+      a2 += to_f32<T>(p[2]); a3 += to_f32<T>(p[3]);
+      b0 += to_f32<T>(q[0]); b1 += to_f32<T>(q[1]);
+      b2 += to_f32<T>(q[2]); b3 += to_f32<T>(q[3]);
+    }
+    if (k < e) {
+      const T* p = data + k * f + fq * 4;
+      a0 += to_f32<T>(p[0]); a1 += to_f32<T>(p[1]);
+      a2 += to_f32<T>(p[2]); a3 += to_f32<T>(p[3]);
+    }
+    float* o = partial + c * f + fq * 4;
+    o[0] = a0 + b0; o[1] = a1 + b1; o[2] = a2 + b2; o[3] = a3 + b3;
+  }
+}
+
 // stage 2: combine of each segment's chunk partials. Block-parallel: one
 // block per (segment, 32-feature tile); 256 threads = 32 features x 8
 // chunk-lanes, fixed-shape LDS tree over the chunk lanes (deterministic).
@@ -328,12 +392,31 @@ torch::Tensor segment_reduce_chunked(torch::Tensor data, torch::Tensor rowptr,
         const T* dp = reinterpret_cast<const T*>(d.data_ptr());
         T* op = reinterpret_cast<T*>(out.data_ptr());
         int threads = 256;
-        long ftiles = (f + threads - 1) / threads;
-        seg_reduce_chunk<T><<<num_blocks(nchunks * ftiles, 1), threads, 0,
-                              stream>>>(dp, cb.data_ptr<long>(),
-                                        ce.data_ptr<long>(),
-                                        partial.data_ptr<float>(), nchunks,
-                                        (int)f);
+        if (f <= 4) {
+          seg_reduce_chunk_small<T, 4>
+              <<<num_blocks(nchunks * WAVE, threads), threads, 0, stream>>>(
+                  dp, cb.data_ptr<long>(), ce.data_ptr<long>(),
+                  partial.data_ptr<float>(), nchunks, (int)f);
+        } else if (f <= 16) {
+          seg_reduce_chunk_small<T, 16>
+              <<<num_blocks(nchunks * WAVE, threads), threads, 0, stream>>>(
+                  dp, cb.data_ptr<long>(), ce.data_ptr<long>(),
+                  partial.data_ptr<float>(), nchunks, (int)f);
+        } else if ((f & 3) == 0) {
+          long ftiles = ((f >> 2) + threads - 1) / threads;
+          seg_reduce_chunk4<T><<<num_blocks(nchunks * ftiles, 1), threads, 0,
+                                 stream>>>(dp, cb.data_ptr<long>(),
+                                           ce.data_ptr<long>(),
+                                           partial.data_ptr<float>(),
+                                           nchunks, (int)f);
+        } else {
+          long ftiles = (f + threads - 1) / threads;
+          seg_reduce_chunk<T><<<num_blocks(nchunks * ftiles, 1), threads, 0,
+                                stream>>>(dp, cb.data_ptr<long>(),
+                                          ce.data_ptr<long>(),
+                                          partial.data_ptr<float>(), nchunks,
+                                          (int)f);
+        }
         long jt = (f + 31) / 32;  // one block per (segment, 32-feat tile)
         seg_reduce_combine<T><<<num_blocks(n * jt * 256, 256), 256, 0,
                                 stream>>>(partial.data_ptr<float>(),

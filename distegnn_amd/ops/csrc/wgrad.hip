@@ -18,9 +18,10 @@ namespace {
 
 constexpr int O_DIM = 64;       // rows of dW (layer width) — fixed
 constexpr int THREADS = 256;    // 4 waves
-constexpr int E_STEP = 32;      // MFMA K per step
-// split-K granularity: chunk chosen at launch for ~1024 blocks
-constexpr int T_STRIDE = 40;    // LDS e-stride (bank-conflict pad)
+// E_STEP: MFMA K per staging round; T_STRIDE = E_STEP + 2 LDS e-stride —
+// (T_STRIDE/2) odd makes the 16 b64 readers of a fragment hit distinct
+// banks. Small-NTW shapes (I=64) are staging-bound and do better with the
+// shorter round; wide shapes (I=144+) amortize staging over 3x the MFMAs.
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
@@ -28,7 +29,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 // NTW = n-tiles (of 16 cols of x) per wave; ITILES total = waves used * NTW
 // mapping: wave w covers n-tiles [w*NTW, w*NTW+NTW)
-template <int NTW>
+template <int NTW, int E_STEP, int T_STRIDE>
 __global__ __launch_bounds__(THREADS) void wgrad_splitk(
     const bf16* __restrict__ g,  // [M, 64]
     const bf16* __restrict__ x,  // [M, I]
@@ -81,18 +82,21 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
     }
     __syncthreads();
     // D[o][i] += gT[o][e] * xT[i][e] — A rows = o, B cols = i, K = e
-    int kb = (lane >> 4) * 8;
 #pragma unroll
-    for (int nt = 0; nt < NTW; ++nt) {
-      int icol = (wave * NTW + nt) * 16 + (lane & 15);
-      if (wave * NTW + nt >= itiles) break;
-      bf16x8 b = *reinterpret_cast<const bf16x8*>(&xT[icol * T_STRIDE + kb]);
+    for (int kk = 0; kk < E_STEP / 32; ++kk) {
+      int kb = kk * 32 + (lane >> 4) * 8;
 #pragma unroll
-      for (int mt = 0; mt < 4; ++mt) {
-        bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            &gT[(mt * 16 + (lane & 15)) * T_STRIDE + kb]);
-        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mt][nt],
-                                                              0, 0, 0);
+      for (int nt = 0; nt < NTW; ++nt) {
+        int icol = (wave * NTW + nt) * 16 + (lane & 15);
+        if (wave * NTW + nt >= itiles) break;
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(&xT[icol * T_STRIDE + kb]);
+#pragma unroll
+        for (int mt = 0; mt < 4; ++mt) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &gT[(mt * 16 + (lane & 15)) * T_STRIDE + kb]);
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[mt][nt], 0, 0, 0);
+        }
       }
     }
   }
@@ -134,7 +138,8 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   auto part = torch::empty({nchunk, (long)O_DIM, (long)ip},
                            g.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
-  int smem = (O_DIM + ip) * T_STRIDE * 2;
+  int t_stride = (((i_dim + 15) / 16 + 3) / 4 >= 3) ? 66 : 34;
+  int smem = (O_DIM + ip) * t_stride * 2;
   int itiles = ip / 16;
   int ntw = (itiles + 3) / 4;
   const bf16* gp = reinterpret_cast<const bf16*>(gc.data_ptr());
@@ -142,20 +147,20 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   float* pp = part.data_ptr<float>();
   switch (ntw) {
     case 1:
-      wgrad_splitk<1><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip, chunk);
+      wgrad_splitk<1, 32, 34><<<(int)nchunk, THREADS, smem, stream>>>(
+          gp, xp, pp, m, i_dim, ip, chunk);
       break;
     case 2:
-      wgrad_splitk<2><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip, chunk);
+      wgrad_splitk<2, 32, 34><<<(int)nchunk, THREADS, smem, stream>>>(
+          gp, xp, pp, m, i_dim, ip, chunk);
       break;
     case 3:
-      wgrad_splitk<3><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip, chunk);
+      wgrad_splitk<3, 64, 66><<<(int)nchunk, THREADS, smem, stream>>>(
+          gp, xp, pp, m, i_dim, ip, chunk);
       break;
     case 4:
-      wgrad_splitk<4><<<(int)nchunk, THREADS, smem, stream>>>(gp, xp, pp, m,
-                                                              i_dim, ip, chunk);
+      wgrad_splitk<4, 64, 66><<<(int)nchunk, THREADS, smem, stream>>>(
+          gp, xp, pp, m, i_dim, ip, chunk);
       break;
     default:
       TORCH_CHECK(false, "unsupported I for wgrad kernel");
