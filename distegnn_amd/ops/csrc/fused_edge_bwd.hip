@@ -59,6 +59,8 @@ struct Smem {
   int bias;     // [4*H] f32
   int wpart;    // [H] f32
   int gbacc;    // [3*H] f32 (block-local gb1/gb2/gb3 column sums)
+  int rows;     // [TILE] i32 (edge row indices, staged once per tile)
+  int cols;     // [TILE] i32
   int total;
 };
 
@@ -74,6 +76,8 @@ __host__ __device__ constexpr Smem smem_layout() {
   L.bias = o; o += 4 * H * 4;
   L.wpart = o; o += H * 4;
   L.gbacc = o; o += 3 * H * 4;
+  L.rows = o; o += TILE * 4;
+  L.cols = o; o += TILE * 4;
   L.total = o;
   return L;
 }
@@ -163,6 +167,13 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
     __syncthreads();
+    int* rws = reinterpret_cast<int*>(smem + L.rows);
+    int* cls = reinterpret_cast<int*>(smem + L.cols);
+    for (int e = tid; e < TILE; e += THREADS) {
+      rws[e] = e < nedge ? (int)row[e0 + e] : 0;
+      cls[e] = e < nedge ? (int)col[e0 + e] : 0;
+    }
+    __syncthreads();
 
     // ---- stage ein ----
     for (int idx = tid; idx < TILE * 16; idx += THREADS) {
@@ -171,8 +182,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       int c8 = (piece & 7) * 8;
       bf16x8 v = {};
       if (e < nedge) {
-        long ge = e0 + e;
-        long src = piece < 8 ? row[ge] : col[ge];
+        long src = piece < 8 ? rws[e] : cls[e];
         v = g8(h + src * H + c8);
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
@@ -183,7 +193,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       float dx = 0, dy = 0, dz = 0, r2 = 0, a0 = 0, a1 = 0;
       if (e < nedge) {
         long ge = e0 + e;
-        long i = row[ge], j = col[ge];
+        long i = rws[e], j = cls[e];
         dx = coord[i * 3] - coord[j * 3];
         dy = coord[i * 3 + 1] - coord[j * 3 + 1];
         dz = coord[i * 3 + 2] - coord[j * 3 + 2];
@@ -294,8 +304,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float dp = 0.f;
       if (e < nedge) {
-        long ge = e0 + e;
-        long i = row[ge];
+        long i = rws[e];
         float inv = normalize ? 1.f / (sqrtf(dptr[3]) + eps) : 1.f;
         dp = (dtrans_n[i * 3] * dptr[0] + dtrans_n[i * 3 + 1] * dptr[1] +
               dtrans_n[i * 3 + 2] * dptr[2]) * inv;
@@ -337,16 +346,25 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       mm_g<2, false>(smem, L.z3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w3t), H, lane, acc);
       __syncthreads();
+      // z3 consumed: reuse its tile to stage dmsg_n[row] COALESCED
+      // (the C-layout merge otherwise issues 16 scattered 2 B loads/lane)
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        int c8 = (idx % 8) * 8;
+        bf16x8 v = {};
+        if (e < nedge) v = g8(dmsg_n + (long)rws[e] * H + c8);
+        *reinterpret_cast<bf16x8*>(smem + L.z3 + (e * H_STRIDE + c8) * 2) = v;
+      }
+      __syncthreads();
       __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
+      const __bf16* dmsg_t = reinterpret_cast<const __bf16*>(smem + L.z3);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int e = wave * 16 + (lane >> 4) * 4 + r;
-          float up = 0.f;
-          if (e < nedge)
-            up = (float)((const __bf16*)dmsg_n)[row[e0 + e] * H + c];
+          float up = (float)dmsg_t[e * H_STRIDE + c];
           float z = (float)z2[e * H_STRIDE + c];
           z2[e * H_STRIDE + c] = (__bf16)((acc[nt][r] + up) * dsilu_(z));
         }
@@ -456,7 +474,7 @@ __global__ __launch_bounds__(THREADS) void fused_edge_bwd(
       long ge = e0 + e;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       const float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
-      long i = row[ge];
+      long i = rws[e];
       float tx = dtrans_n[i * 3], ty = dtrans_n[i * 3 + 1],
             tz = dtrans_n[i * 3 + 2];
       float p = sc[0], dr2 = sc[2];
