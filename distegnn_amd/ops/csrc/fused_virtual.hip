@@ -387,20 +387,37 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     }
     __syncthreads();
 
-    // dzxv = dpxv * wxv o silu'(zxv); dzX likewise (elementwise loads)
+    // dzxv = dpxv * wxv o silu'(zxv); dzX likewise.
+    // zxv/zX are staged COALESCED through the (currently free) za and
+    // in_tile regions — the C-layout loop otherwise issues 16 scattered
+    // 2 B global loads per lane per tensor.
+    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+      int e = idx / 8;
+      int c8 = (idx % 8) * 8;
+      bf16x8 v = {}, w = {};
+      if (e < nrow) {
+        v = g8(zxv_in + (r0 + e) * H + c8);
+        w = g8(zX_in + (r0 + e) * H + c8);
+      }
+      *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
+      *reinterpret_cast<bf16x8*>(smem + L.in_tile
+                                 + (e * H_STRIDE + c8) * 2) = w;
+    }
+    __syncthreads();
     {
       int cc = tid & 63;
       int estart = (tid >> 6) * 16;
       float wv0 = biases[4 * H + cc], wv1 = biases[5 * H + cc];
       __bf16* zc = reinterpret_cast<__bf16*>(smem + L.zc);
       __bf16* zd = reinterpret_cast<__bf16*>(smem + L.zd);
+      const __bf16* zxs = reinterpret_cast<const __bf16*>(smem + L.za);
+      const __bf16* zXs = reinterpret_cast<const __bf16*>(smem + L.in_tile);
       const float* sc = reinterpret_cast<const float*>(smem + L.scal);
       float abxv = 0.f, abX = 0.f, awxvv = 0.f, awXv = 0.f;
       for (int e = estart; e < estart + 16; ++e) {
-        long r = r0 + e;
         bool ok = e < nrow;
-        float zx = ok ? (float)((const __bf16*)zxv_in)[r * H + cc] : 0.f;
-        float zX_ = ok ? (float)((const __bf16*)zX_in)[r * H + cc] : 0.f;
+        float zx = (float)zxs[e * H_STRIDE + cc];
+        float zX_ = (float)zXs[e * H_STRIDE + cc];
         float dpxv = sc[e * 4], dpX = sc[e * 4 + 1];
         float dxv = dpxv * wv0 * dsilu_(zx);
         float dX = dpX * wv1 * dsilu_(zX_);
@@ -424,24 +441,36 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
 
     // dvmsg_tot = dvmsg + dzxv@Wxv + dzX@WX; dz2 = dvmsg_tot o silu'(z2)
     {
+      // re-stage za/in_tile (consumed above) with dvmsg and z2, coalesced
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        int c8 = (idx % 8) * 8;
+        bf16x8 v = {}, w = {};
+        if (e < nrow) {
+          v = g8(dvmsg + (r0 + e) * H + c8);
+          w = g8(z2_in + (r0 + e) * H + c8);
+        }
+        *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
+        *reinterpret_cast<bf16x8*>(smem + L.in_tile
+                                   + (e * H_STRIDE + c8) * 2) = w;
+      }
       f32x4 acc[4] = {};
       mm_g<2, false>(smem, L.zc + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(wxvt), H, lane, acc);
       mm_g<2, false>(smem, L.zd + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(wXt), H, lane, acc);
+      __syncthreads();
       __bf16* zb = reinterpret_cast<__bf16*>(smem + L.zb);
+      const __bf16* ups = reinterpret_cast<const __bf16*>(smem + L.za);
+      const __bf16* z2s = reinterpret_cast<const __bf16*>(smem + L.in_tile);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int cc = nt * 16 + (lane & 15);
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) {
           int e = wave * 16 + (lane >> 4) * 4 + rr;
-          long r = r0 + e;
-          float up = 0.f, z2v = 0.f;
-          if (e < nrow) {
-            up = (float)((const __bf16*)dvmsg)[r * H + cc];
-            z2v = (float)((const __bf16*)z2_in)[r * H + cc];
-          }
+          float up = (float)ups[e * H_STRIDE + cc];
+          float z2v = (float)z2s[e * H_STRIDE + cc];
           zb[e * H_STRIDE + cc] =
               (__bf16)((acc[nt][rr] + up) * dsilu_(z2v));
         }
@@ -459,11 +488,20 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       atomicAdd(&gbacc[H + cc], ab);
     }
 
-    // dz1 = (dz2 @ W2) o silu'(z1)
+    // dz1 = (dz2 @ W2) o silu'(z1); z1 staged into za coalesced, each
+    // slot read once then overwritten in place by its dz1 value
     {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        int c8 = (idx % 8) * 8;
+        bf16x8 v = {};
+        if (e < nrow) v = g8(z1_in + (r0 + e) * H + c8);
+        *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
+      }
       f32x4 acc[4] = {};
       mm_g<2, false>(smem, L.zb + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w2t), H, lane, acc);
+      __syncthreads();
       __bf16* za = reinterpret_cast<__bf16*>(smem + L.za);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -471,9 +509,7 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) {
           int e = wave * 16 + (lane >> 4) * 4 + rr;
-          long r = r0 + e;
-          float z1v = e < nrow
-              ? (float)((const __bf16*)z1_in)[r * H + cc] : 0.f;
+          float z1v = (float)za[e * H_STRIDE + cc];
           za[e * H_STRIDE + cc] = (__bf16)(acc[nt][rr] * dsilu_(z1v));
         }
       }
