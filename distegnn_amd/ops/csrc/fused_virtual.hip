@@ -85,15 +85,17 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
 }
 
 struct VSmem {
-  int in_tile;  // [TILE][K_STRIDE] bf16
-  int za;       // [TILE][H_STRIDE] bf16 (z1; bwd: scratch/dz1)
+  int in_tile;  // [TILE][K_STRIDE] bf16 (bwd also time-shares it for the
+                //   zX/z2 stagings and the dzX tile -> 3 blocks/CU)
+  int za;       // [TILE][H_STRIDE] bf16 (z1; bwd: stagings/dz1)
   int zb;       // [TILE][H_STRIDE] bf16 (z2; bwd: dz2)
-  int zc;       // [TILE][H_STRIDE] bf16 (zxv / zX time-shared; bwd: dzxv/dzX)
-  int zd;       // [TILE][H_STRIDE] bf16 (bwd: second head tile)
+  int zc;       // [TILE][H_STRIDE] bf16 (zxv / zX time-shared; bwd: dzxv)
   int diff;     // [TILE][4] f32 (vdiff xyz, vrad)
   int scal;     // [TILE][4] f32 (pxv, pX / dpxv, dr)
   int bias;     // [6*H] f32 (b1, b2, bxv, bX, wxv, wX)
   int gbacc;    // [6*H] f32 (bwd: gb1, gb2, gbxv, gbX, gwxvv, gwXv)
+  int zd;       // end marker: kernels allocate smem up to here (zd itself
+                //   is no longer a live region)
   int total;
 };
 
@@ -104,11 +106,11 @@ __host__ __device__ constexpr VSmem vsmem_layout() {
   L.za = o; o += TILE * H_STRIDE * 2;
   L.zb = o; o += TILE * H_STRIDE * 2;
   L.zc = o; o += TILE * H_STRIDE * 2;
-  L.zd = o; o += TILE * H_STRIDE * 2;
   L.diff = o; o += TILE * 4 * 4;
   L.scal = o; o += TILE * 4 * 4;
   L.bias = o; o += 6 * H * 4;
   L.gbacc = o; o += 6 * H * 4;
+  L.zd = o;
   L.total = o;
   return L;
 }
@@ -409,9 +411,10 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       int estart = (tid >> 6) * 16;
       float wv0 = biases[4 * H + cc], wv1 = biases[5 * H + cc];
       __bf16* zc = reinterpret_cast<__bf16*>(smem + L.zc);
-      __bf16* zd = reinterpret_cast<__bf16*>(smem + L.zd);
+      // dzX overwrites the zX staging in place (same slot, same thread)
+      __bf16* zd = reinterpret_cast<__bf16*>(smem + L.in_tile);
       const __bf16* zxs = reinterpret_cast<const __bf16*>(smem + L.za);
-      const __bf16* zXs = reinterpret_cast<const __bf16*>(smem + L.in_tile);
+      const __bf16* zXs = zd;
       const float* sc = reinterpret_cast<const float*>(smem + L.scal);
       float abxv = 0.f, abX = 0.f, awxvv = 0.f, awXv = 0.f;
       for (int e = estart; e < estart + 16; ++e) {
@@ -437,11 +440,17 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     }
     __syncthreads();
     TILE_TO_GLOBAL(L.zc, dzxv_out, false);
-    TILE_TO_GLOBAL(L.zd, dzX_out, false);
+    TILE_TO_GLOBAL(L.in_tile, dzX_out, false);
 
     // dvmsg_tot = dvmsg + dzxv@Wxv + dzX@WX; dz2 = dvmsg_tot o silu'(z2)
     {
-      // re-stage za/in_tile (consumed above) with dvmsg and z2, coalesced
+      f32x4 acc[4] = {};
+      mm_g<2, false>(smem, L.zc + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+                     opaque(wxvt), H, lane, acc);
+      mm_g<2, false>(smem, L.in_tile + wave * 16 * H_STRIDE * 2,
+                     H_STRIDE * 2, opaque(wXt), H, lane, acc);
+      __syncthreads();
+      // in_tile consumed: re-stage za/in_tile with dvmsg and z2, coalesced
       for (int idx = tid; idx < TILE * 8; idx += THREADS) {
         int e = idx / 8;
         int c8 = (idx % 8) * 8;
@@ -454,11 +463,6 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
         *reinterpret_cast<bf16x8*>(smem + L.in_tile
                                    + (e * H_STRIDE + c8) * 2) = w;
       }
-      f32x4 acc[4] = {};
-      mm_g<2, false>(smem, L.zc + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                     opaque(wxvt), H, lane, acc);
-      mm_g<2, false>(smem, L.zd + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                     opaque(wXt), H, lane, acc);
       __syncthreads();
       __bf16* zb = reinterpret_cast<__bf16*>(smem + L.zb);
       const __bf16* ups = reinterpret_cast<const __bf16*>(smem + L.za);
@@ -667,9 +671,9 @@ std::vector<torch::Tensor> fused_virtual_forward(
       cdim, k_in
 
   if (train) {
-    fused_virtual_fwd<true><<<blocks, THREADS, L.total, stream>>>(ARGS);
+    fused_virtual_fwd<true><<<blocks, THREADS, L.zd, stream>>>(ARGS);
   } else {
-    fused_virtual_fwd<false><<<blocks, THREADS, L.total, stream>>>(ARGS);
+    fused_virtual_fwd<false><<<blocks, THREADS, L.zd, stream>>>(ARGS);
   }
 #undef ARGS
   return {vmsg, tv, tx, vin, z1, z2, zxv, zX, p2};
@@ -712,7 +716,7 @@ std::vector<torch::Tensor> fused_virtual_backward(
   auto wXvc = wXv.contiguous().to(torch::kFloat);
   auto cc_ = coord.contiguous().to(torch::kFloat);
   auto vc = vcoord.contiguous().to(torch::kFloat);
-  fused_virtual_bwd<<<blocks, THREADS, L.total, stream>>>(
+  fused_virtual_bwd<<<blocks, THREADS, L.zd, stream>>>(
       cc_.data_ptr<float>(), vc.data_ptr<float>(),
       batch.contiguous().data_ptr<long>(),
       reinterpret_cast<const bf16*>(dvmsg.contiguous().data_ptr()),
