@@ -138,7 +138,7 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   auto part = torch::empty({nchunk, (long)O_DIM, (long)ip},
                            g.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
-  int t_stride = (((i_dim + 15) / 16 + 3) / 4 >= 3) ? 66 : 34;
+  int t_stride = (((i_dim + 15) / 16 + 3) / 4 >= 3) ? 66 : 40;
   int smem = (O_DIM + ip) * t_stride * 2;
   int itiles = ip / 16;
   int ntw = (itiles + 3) / 4;
@@ -147,11 +147,11 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
   float* pp = part.data_ptr<float>();
   switch (ntw) {
     case 1:
-      wgrad_splitk<1, 32, 34><<<(int)nchunk, THREADS, smem, stream>>>(
+      wgrad_splitk<1, 32, 40><<<(int)nchunk, THREADS, smem, stream>>>(
           gp, xp, pp, m, i_dim, ip, chunk);
       break;
     case 2:
-      wgrad_splitk<2, 32, 34><<<(int)nchunk, THREADS, smem, stream>>>(
+      wgrad_splitk<2, 32, 40><<<(int)nchunk, THREADS, smem, stream>>>(
           gp, xp, pp, m, i_dim, ip, chunk);
       break;
     case 3:
