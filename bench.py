@@ -179,6 +179,12 @@ def main():
     if world_size != args.gpus and int(os.environ.get("WORLD_SIZE", "1")) > 1:
         args.gpus = world_size
     world_size = max(world_size, 1)
+    # DISTEGNN_FORCE_DIST=1 under torchrun -nproc 1: run the FULL collective
+    # path (RCCL all-reduces, GradBucket) on a 1-rank group — identity math,
+    # but exercises RCCL capture inside hipGraphs on a single GPU.
+    force_dist = (os.environ.get("DISTEGNN_FORCE_DIST") == "1"
+                  and world_size == 1 and comm.is_distributed())
+    ws_eff = 2 if force_dist else world_size
     assert torch.cuda.is_available(), "bench.py requires a GPU"
     device = torch.device(f"cuda:{rank}")
     torch.cuda.set_device(device)
@@ -207,10 +213,10 @@ def main():
 
     model = FastEGNN(node_feat_nf=feat_nf, node_attr_nf=attr_nf,
                      edge_attr_nf=2, hidden_nf=64, virtual_channels=vch,
-                     world_size=world_size, n_layers=4,
+                     world_size=ws_eff, n_layers=4,
                      normalize=normalize).to(device)
     grad_bucket = None
-    if world_size > 1:
+    if ws_eff > 1:
         grad_bucket = GradBucket(model)
         grad_bucket.broadcast_parameters()
     optimizer = torch.optim.Adam(model.parameters(), lr=5e-4,
@@ -233,7 +239,7 @@ def main():
     mse = None
     for w in range(args.warmup):
         mse = train_step(graphed, batches[w % len(batches)], optimizer,
-                         grad_bucket, w, accum, world_size, device, clip,
+                         grad_bucket, w, accum, ws_eff, device, clip,
                          mmd_cfg=num_sample)
     comm.barrier()
     torch.cuda.synchronize()
@@ -241,11 +247,11 @@ def main():
     for k in range(args.steps):
         mse = train_step(graphed, batches[k % len(batches)], optimizer,
                          grad_bucket, args.warmup + k, accum,
-                         world_size, device, clip, mmd_cfg=num_sample)
+                         ws_eff, device, clip, mmd_cfg=num_sample)
     comm.barrier()
     torch.cuda.synchronize()
     elapsed = torch.tensor(time.perf_counter() - t0, device=device)
-    if world_size > 1:
+    if ws_eff > 1:
         torch.distributed.all_reduce(elapsed,
                                      op=torch.distributed.ReduceOp.MAX)
     ms_per_step = elapsed.item() * 1000.0 / args.steps

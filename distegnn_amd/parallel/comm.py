@@ -43,7 +43,8 @@ def init_distributed(backend: Optional[str] = None,
     CUDA devices are visible, else gloo.
     """
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
-    if world_size <= 1:
+    force = os.environ.get("DISTEGNN_FORCE_DIST") == "1"
+    if world_size <= 1 and not (force and "RANK" in os.environ):
         return 0, 1
     local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
     if backend is None:
