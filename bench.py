@@ -169,10 +169,12 @@ def main():
                     help="distinct synthetic samples to cycle through")
     ap.add_argument("--graphs", type=str, default="auto",
                     choices=["auto", "on", "off"],
-                    help="hipGraph-capture the fwd+loss+bwd step; auto = on "
-                         "for 1 GPU, off for multi-GPU (RCCL capture of the "
-                         "in-forward virtual-node all-reduce is untested on "
-                         "this pool; force with --graphs on)")
+                    help="hipGraph-capture the fwd+loss+bwd step (auto = on; "
+                         "RCCL capture of the in-forward virtual-node "
+                         "all-reduce is validated on a 1-rank group via "
+                         "DISTEGNN_FORCE_DIST=1; asymmetric capture failure "
+                         "is deadlock-free because the eager fallback issues "
+                         "the same collectives in the same order)")
     args = ap.parse_args()
 
     rank, world_size = comm.init_distributed()
@@ -228,8 +230,7 @@ def main():
 
     step_core = make_step_core(model, accum, mmd_sigma, mmd_samples,
                                autocast_dtype)
-    use_graphs = (args.graphs == "on"
-                  or (args.graphs == "auto" and world_size == 1))
+    use_graphs = args.graphs != "off"
     graphed = GraphedStep(step_core, model.parameters(),
                           warmup_occurrences=2,
                           enabled=use_graphs, verbose=True)
