@@ -119,6 +119,27 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
   (void)ntw_used;
 }
 
+// Row-streaming partial combine: each block owns (row-chunk, col-tile) and
+// sums its rows with coalesced reads; a second pass folds the row-chunk
+// partials. Fixed tree order -> deterministic. aten's sum(0) walks the
+// [nchunk, 64*ip] array column-major (~700 GB/s for this shape).
+__global__ void colsum_rows(const float* __restrict__ src,
+                            float* __restrict__ dst, long rows, long cols,
+                            long rows_per_chunk) {
+  long nchunks = (rows + rows_per_chunk - 1) / rows_per_chunk;
+  long ctiles = (cols + 255) / 256;
+  for (long b = blockIdx.x; b < nchunks * ctiles; b += gridDim.x) {
+    long rc = b / ctiles;
+    long c = (b - rc * ctiles) * 256 + threadIdx.x;
+    if (c >= cols) continue;
+    long r0 = rc * rows_per_chunk;
+    long r1 = r0 + rows_per_chunk < rows ? r0 + rows_per_chunk : rows;
+    float acc = 0.f;
+    for (long r = r0; r < r1; ++r) acc += src[r * cols + c];
+    dst[rc * cols + c] = acc;
+  }
+}
+
 }  // namespace
 
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
@@ -165,6 +186,22 @@ torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x) {
     default:
       TORCH_CHECK(false, "unsupported I for wgrad kernel");
   }
-  auto dw = part.sum(0);  // [64, ip]
+  // two-pass row-streaming combine of the [nchunk, 64*ip] partials
+  long cols = (long)O_DIM * ip;
+  auto dw = torch::empty({(long)O_DIM, (long)ip},
+                         g.options().dtype(torch::kFloat));
+  if (nchunk <= 16) {
+    colsum_rows<<<num_blocks(((cols + 255) / 256) * 256, 256), 256, 0,
+                  stream>>>(pp, dw.data_ptr<float>(), nchunk, cols, nchunk);
+  } else {
+    long rpc = 16;
+    long nmid = (nchunk + rpc - 1) / rpc;
+    auto mid = torch::empty({nmid, cols}, g.options().dtype(torch::kFloat));
+    colsum_rows<<<num_blocks(nmid * ((cols + 255) / 256) * 256, 256), 256, 0,
+                  stream>>>(pp, mid.data_ptr<float>(), nchunk, cols, rpc);
+    colsum_rows<<<num_blocks(((cols + 255) / 256) * 256, 256), 256, 0,
+                  stream>>>(mid.data_ptr<float>(), dw.data_ptr<float>(),
+                            nmid, cols, nmid);
+  }
   return dw.narrow(1, 0, i_dim);
 }
