@@ -43,7 +43,6 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int itiles = ip / 16;
-  const int ntw_used = (itiles + 3) / 4;  // n-tiles this wave actually owns
 
   long c0 = (long)blockIdx.x * chunk;
   long c1 = c0 + chunk < m ? c0 + chunk : m;
@@ -116,7 +115,6 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
       }
     }
   }
-  (void)ntw_used;
 }
 
 // Row-streaming partial combine: each block owns (row-chunk, col-tile) and

@@ -77,7 +77,6 @@ class GraphedStep:
         self.enabled = enabled and torch.cuda.is_available()
         self.verbose = verbose
         self.entries: Dict[Tuple, _ShapeEntry] = {}
-        self._pool = None  # shared mempool across shape keys (see capture)
         self._side = None  # side stream for eager work between replays
 
     def run_eager(self, fn):
@@ -165,7 +164,7 @@ class GraphedStep:
                              for (shape, dtype, dev) in e.out_meta)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g, pool=self._pool):
+            with torch.cuda.graph(g):
                 outputs = self.step_fn(proxy)
                 for buf, o in zip(out_bufs, outputs):
                     buf.copy_(o)

@@ -41,9 +41,6 @@ def run(graphs_on, steps=24):
                   f"{em:.7f}")
     return out
 
-import os
-if "--shared-pool" in sys.argv:
-    os.environ["DISTEGNN_GRAPHS_SHARED_POOL"] = "1"
 import distegnn_amd.runtime.losses as L
 import bench as B
 
