@@ -30,10 +30,10 @@ Design:
   replays — loss scaled wrong, params eventually NaN). ``run_eager``
   runs the callable on a dedicated side stream whose per-stream blocks
   never overlap the graph pool.
-* Distributed: RCCL collectives inside graphs are gated by
-  ``allow_collectives`` (hipGraph capture of RCCL is exercised by the
-  multi-GPU bench path; on failure we fall back to eager permanently for
-  that shape).
+* Distributed: the virtual-node RCCL all-reduces are captured INSIDE the
+  graph (validated on a 1-rank group via bench.py DISTEGNN_FORCE_DIST=1).
+  Asymmetric capture failure cannot deadlock ranks: the eager fallback
+  re-issues the same collectives in the same program order.
 
 Limitations: replay requires the SAME dtypes/fields per shape key; any
 capture failure disables capture for that key and logs once.
