@@ -285,6 +285,27 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
     log_dict = {"epochs": [], "loss": [], "loss_train": []}
     best_log_dict = {"epoch_index": 0, "loss_valid": 1e8, "loss_test": 1e8,
                      "loss_train": 1e8}
+    # wandb (rank 0, reference utils/train.py:185-198): enabled by config,
+    # offline by default; package absence degrades to a one-line warning.
+    wb = None
+    wcfg = getattr(log_config, "wandb", None)
+    if rank == 0 and wcfg is not None and getattr(wcfg, "enable", False):
+        try:
+            import wandb as wb  # type: ignore
+
+            if getattr(wcfg, "offline", True):
+                os.environ.setdefault("WANDB_MODE", "offline")
+            if getattr(wcfg, "api_key", ""):
+                os.environ.setdefault("WANDB_API_KEY", wcfg.api_key)
+            wb.init(project=getattr(wcfg, "project", "") or None,
+                    entity=getattr(wcfg, "entity", "") or None,
+                    name=log_config.exp_name,
+                    config=config.to_dict() if hasattr(config, "to_dict")
+                    else dict(config))
+        except ImportError:
+            print("[trainer] wandb enabled in config but not installed; "
+                  "continuing without it")
+            wb = None
     if rank == 0:
         log_dir = os.path.join(log_config.log_dir, log_config.exp_name, "log")
         os.makedirs(log_dir, exist_ok=True)
@@ -351,6 +372,9 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                 torch.save(state, os.path.join(state_dict_dir,
                                                "last_model.pth"))
 
+            if rank == 0 and wb is not None:
+                wb.log({"loss_train": loss_train, "loss_valid": loss_valid,
+                        "loss_test": loss_test, "epoch": epoch_index})
             if rank == 0 and (epoch_index - best_log_dict["epoch_index"]
                               >= train_config.early_stop):
                 best_log_dict["early_stop"] = epoch_index
@@ -371,4 +395,7 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
         dist.barrier()
         comm.destroy()
     if rank == 0:
+        if wb is not None:
+            wb.log({"best_test_loss": best_log_dict["loss_test"]})
+            wb.finish()
         return best_log_dict, log_dict
