@@ -208,7 +208,7 @@ class EGCLVel(nn.Module):
                 self.coord_mlp_v_virtual[0].weight,
                 self.coord_mlp_v_virtual[0].bias,
                 self.coord_mlp_v_virtual[2].weight.reshape(-1))
-            trans_v = tv.mean(dim=1)
+            trans_v = ops.mid_mean(tv)
         else:
             vdiff = (ops.gather_rows(virtual_coord, batch, ptr,
                                      chunks=pool_chunks)
@@ -259,7 +259,7 @@ class EGCLVel(nn.Module):
         # --- node model, phi_h ------------------------------------------
         if not fuse:
             agg_e = ops.segment_mean(edge_feat, row, n, rowptr=rowptr)
-        agg_v = v_msg.mean(dim=1)
+        agg_v = ops.mid_mean(v_msg)
         if node_attr is not None:
             node_in = torch.cat([h, agg_e, agg_v, node_attr], dim=1)
         else:

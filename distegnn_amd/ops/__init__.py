@@ -164,6 +164,33 @@ def graph_mean_pool(x: torch.Tensor, batch: torch.Tensor, num_graphs: int,
     return reference.graph_mean_pool(x, batch, num_graphs, counts=counts)
 
 
+class _MidMeanFn(torch.autograd.Function):
+    """mean over the middle (channel) dim of [N, C, F] — one HIP kernel
+    each way instead of aten reduce + broadcast launches."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ext = _require_ext("mid_mean")
+        ctx.c = x.size(1)
+        if ext is None:
+            return x.mean(dim=1)
+        return ext.mid_reduce(x.contiguous(), 1.0 / ctx.c)
+
+    @staticmethod
+    def backward(ctx, gout):
+        ext = _load_extension()
+        if ext is None or not gout.is_cuda:
+            return gout.unsqueeze(1).expand(-1, ctx.c, -1) / ctx.c
+        return ext.mid_expand(gout.contiguous(), ctx.c, 1.0 / ctx.c)
+
+
+def mid_mean(x: torch.Tensor) -> torch.Tensor:
+    """x.mean(dim=1) for [N, C, F] (HIP-fused on GPU)."""
+    if x.is_cuda and x.dim() == 3:
+        return _MidMeanFn.apply(x)
+    return x.mean(dim=1)
+
+
 class _GatherRowsFn(torch.autograd.Function):
     """index_select(0, idx) whose BACKWARD is a deterministic CSR segment
     sum instead of torch's index_add scatter.
@@ -444,8 +471,8 @@ class _FusedVirtualBlockFn(torch.autograd.Function):
         gbxv, gbX = gb[128:192], gb[192:256]
         gwxvv, gwXv = gb[256:320], gb[320:384]
 
-        gh = dh_row.view(n, c, -1).sum(1, dtype=torch.float32).to(h.dtype)
-        gcoord = -dvd.view(n, c, 3).sum(1)
+        gh = ext.mid_reduce(dh_row.view(n, c, -1), 1.0)
+        gcoord = ext.mid_reduce(dvd.view(n, c, 3), -1.0)
         # per-graph pools (sum) of the [B,C,*] gradients
         def pool(x_rows, width):
             flat = x_rows.reshape(n, c * width)

@@ -7,6 +7,8 @@ torch::Tensor segment_reduce_csr_perm(torch::Tensor data,
                                       torch::Tensor rowptr,
                                       torch::Tensor perm, bool mean);
 torch::Tensor gather_rows_fast(torch::Tensor data, torch::Tensor idx);
+torch::Tensor mid_reduce(torch::Tensor x, double scale);
+torch::Tensor mid_expand(torch::Tensor g, int64_t c, double scale);
 torch::Tensor segment_reduce_chunked(torch::Tensor data, torch::Tensor rowptr,
                                      torch::Tensor chunk_begin,
                                      torch::Tensor chunk_end,
@@ -56,6 +58,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows_fast", &gather_rows_fast,
         "vectorized dst[i] = src[idx[i]] row gather", py::arg("data"),
         py::arg("idx"));
+  m.def("mid_reduce", &mid_reduce,
+        "scale * sum over the middle dim of [N,C,F]", py::arg("x"),
+        py::arg("scale"));
+  m.def("mid_expand", &mid_expand,
+        "broadcast [N,F] over a middle dim: [N,C,F] * scale", py::arg("g"),
+        py::arg("c"), py::arg("scale"));
   m.def("segment_reduce_chunked", &segment_reduce_chunked,
         "two-stage deterministic segmented reduce for huge segments",
         py::arg("data"), py::arg("rowptr"), py::arg("chunk_begin"),

@@ -426,3 +426,79 @@ torch::Tensor segment_reduce_chunked(torch::Tensor data, torch::Tensor rowptr,
       });
   return out;
 }
+
+namespace {
+
+// Middle-dimension reduction: x [N, C, F] -> out [N, F], out = scale *
+// sum_c x[:, c, :]. Covers the virtual-channel means/sums (trans_v,
+// agg_v, dh/dcoord channel folds) that otherwise run as ~4 us aten
+// reduce_kernel launches each. fp32 accumulation; scale = 1/C for mean,
+// -1 for negated sums.
+template <typename T>
+__global__ void mid_reduce_kernel(const T* __restrict__ x,
+                                  T* __restrict__ out, long n, int c, int f,
+                                  float scale) {
+  long total = n * f;
+  for (long o = blockIdx.x * (long)blockDim.x + threadIdx.x; o < total;
+       o += (long)gridDim.x * blockDim.x) {
+    long ln = o / f;
+    int j = (int)(o - ln * f);
+    const T* p = x + ln * (long)c * f + j;
+    float acc = 0.f;
+    for (int cc = 0; cc < c; ++cc) acc += to_f32<T>(p[cc * (long)f]);
+    out[o] = from_f32<T>(acc * scale);
+  }
+}
+
+// Broadcast backward of mid_reduce: g [N, F] -> out [N, C, F], scaled.
+template <typename T>
+__global__ void mid_expand_kernel(const T* __restrict__ g,
+                                  T* __restrict__ out, long n, int c, int f,
+                                  float scale) {
+  long total = n * (long)c * f;
+  long cf = (long)c * f;
+  for (long o = blockIdx.x * (long)blockDim.x + threadIdx.x; o < total;
+       o += (long)gridDim.x * blockDim.x) {
+    long ln = o / cf;
+    int j = (int)((o - ln * cf) % f);
+    out[o] = from_f32<T>(to_f32<T>(g[ln * (long)f + j]) * scale);
+  }
+}
+
+}  // namespace
+
+torch::Tensor mid_reduce(torch::Tensor x, double scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3, "mid_reduce: [N,C,F] CUDA");
+  auto d = x.contiguous();
+  long n = d.size(0), c = d.size(1), f = d.size(2);
+  auto out = torch::empty({n, f}, d.options());
+  if (n == 0 || f == 0) return out;
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND(
+      at::ScalarType::BFloat16, d.scalar_type(), "mid_reduce", [&] {
+        using T = typename hip_type<scalar_t>::type;
+        mid_reduce_kernel<T><<<num_blocks(n * f, 256), 256, 0, stream>>>(
+            reinterpret_cast<const T*>(d.data_ptr()),
+            reinterpret_cast<T*>(out.data_ptr()), n, (int)c, (int)f,
+            (float)scale);
+      });
+  return out;
+}
+
+torch::Tensor mid_expand(torch::Tensor g, int64_t c, double scale) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 2, "mid_expand: [N,F] CUDA");
+  auto d = g.contiguous();
+  long n = d.size(0), f = d.size(1);
+  auto out = torch::empty({n, c, f}, d.options());
+  if (n == 0 || f == 0 || c == 0) return out;
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND(
+      at::ScalarType::BFloat16, d.scalar_type(), "mid_expand", [&] {
+        using T = typename hip_type<scalar_t>::type;
+        mid_expand_kernel<T><<<num_blocks(n * c * f, 256), 256, 0, stream>>>(
+            reinterpret_cast<const T*>(d.data_ptr()),
+            reinterpret_cast<T*>(out.data_ptr()), n, (int)c, (int)f,
+            (float)scale);
+      });
+  return out;
+}
