@@ -171,3 +171,45 @@ def test_segment_autograd_gpu():
     out_g.pow(2).sum().backward()
     assert torch.allclose(out_g.cpu(), out, atol=1e-4, rtol=1e-4)
     assert torch.allclose(data_g.grad.cpu(), data.grad, atol=1e-4, rtol=1e-4)
+
+
+def test_mid_reduce_expand():
+    from distegnn_amd.ops import mid_mean, hip_ext
+    x = torch.randn(500, 5, 64, device="cuda:0", dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = mid_mean(x)
+    ref = x.detach().float().mean(dim=1).to(torch.bfloat16)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gref = (g.float() / 5).unsqueeze(1).expand(-1, 5, -1)
+    assert torch.allclose(x.grad.float(), gref, atol=2e-2)
+    ext = hip_ext()
+    y = torch.randn(300, 3, 3, device="cuda:0")
+    assert torch.allclose(ext.mid_reduce(y, -1.0), -y.sum(1), atol=1e-5)
+
+
+def test_gather_rows_fast_matches_index_select():
+    from distegnn_amd.ops import hip_ext
+    ext = hip_ext()
+    for dtype, f in [(torch.bfloat16, 64), (torch.float32, 3),
+                     (torch.float32, 15)]:
+        x = torch.randn(1000, f, device="cuda:0", dtype=dtype)
+        idx = torch.randint(0, 1000, (5000,), device="cuda:0")
+        assert torch.equal(ext.gather_rows_fast(x, idx),
+                           x.index_select(0, idx))
+
+
+def test_segment_reduce_perm():
+    from distegnn_amd.ops import hip_ext
+    ext = hip_ext()
+    m, n, f = 4000, 100, 64
+    data = torch.randn(m, f, device="cuda:0", dtype=torch.bfloat16)
+    seg = torch.sort(torch.randint(0, n, (m,), device="cuda:0")).values
+    rowptr = torch.searchsorted(
+        seg, torch.arange(n + 1, device="cuda:0"), right=False)
+    rowptr[-1] = m
+    perm = torch.randperm(m, device="cuda:0")
+    out = ext.segment_reduce_csr_perm(data, rowptr, perm, False)
+    ref = ext.segment_reduce_csr(data.index_select(0, perm), rowptr, False)
+    assert torch.allclose(out.float(), ref.float(), atol=1e-2, rtol=1e-2)
