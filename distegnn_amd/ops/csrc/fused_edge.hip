@@ -49,6 +49,8 @@ struct SmemLayout {
   int diff;     // [TILE][4] f32
   int pvec;     // [TILE] f32
   int bias;     // [4*H] f32
+  int rows;     // [TILE] i32 (edge endpoints, staged once per tile)
+  int cols;     // [TILE] i32
   int total;
 };
 
@@ -61,6 +63,8 @@ __host__ __device__ constexpr SmemLayout smem_layout() {
   L.diff = o; o += TILE * 4 * 4;
   L.pvec = o; o += TILE * 4;
   L.bias = o; o += 4 * H * 4;
+  L.rows = o; o += TILE * 4;
+  L.cols = o; o += TILE * 4;
   L.total = o;
   return L;
 }
@@ -124,6 +128,13 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
     __syncthreads();
+    int* rws = reinterpret_cast<int*>(smem + L.rows);
+    int* cls = reinterpret_cast<int*>(smem + L.cols);
+    for (int e = tid; e < TILE; e += THREADS) {
+      rws[e] = e < nedge ? (int)row[e0 + e] : 0;
+      cls[e] = e < nedge ? (int)col[e0 + e] : 0;
+    }
+    __syncthreads();
 
     // gather stage: in_tile [64][K_STRIDE]
     for (int idx = tid; idx < TILE * 16; idx += THREADS) {
@@ -132,8 +143,7 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       int c8 = (piece & 7) * 8;
       bf16x8 v = {};
       if (e < nedge) {
-        long ge = e0 + e;
-        long src = piece < 8 ? row[ge] : col[ge];
+        long src = piece < 8 ? rws[e] : cls[e];
         v = g8(h + src * H + c8);
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
@@ -144,7 +154,7 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       float dx = 0, dy = 0, dz = 0, r2 = 0, a0 = 0, a1 = 0;
       if (e < nedge) {
         long ge = e0 + e;
-        long i = row[ge], j = col[ge];
+        long i = rws[e], j = cls[e];
         dx = coord[i * 3] - coord[j * 3];
         dy = coord[i * 3 + 1] - coord[j * 3 + 1];
         dz = coord[i * 3 + 2] - coord[j * 3 + 2];
