@@ -102,14 +102,20 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
 def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
                world_size, device, clip=True, mmd_cfg=None):
     data = batch.to(device)
-    data.counts_global = (comm.global_counts(data.counts)
-                          if world_size > 1 else data.counts)
-    if mmd_cfg is not None:
-        # fresh randomness drawn OUTSIDE the captured region
-        from distegnn_amd.runtime.losses import draw_sample_indices
 
-        data.mmd_idx, data.mmd_valid = draw_sample_indices(
-            data.batch, data.ptr, data.counts, mmd_cfg)
+    def _pre():
+        # eager work between replays runs on the side stream: its
+        # allocations must not alias captured-graph pool blocks
+        data.counts_global = (comm.global_counts(data.counts)
+                              if world_size > 1 else data.counts)
+        if mmd_cfg is not None:
+            # fresh randomness drawn OUTSIDE the captured region
+            from distegnn_amd.runtime.losses import draw_sample_indices
+
+            data.mmd_idx, data.mmd_valid = draw_sample_indices(
+                data.batch, data.ptr, data.counts, mmd_cfg)
+
+    graphed.run_eager(_pre)
     (mse_log,) = graphed(data)
     if (step + 1) % accum == 0:
         if grad_bucket is not None:

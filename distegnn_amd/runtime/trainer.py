@@ -182,13 +182,17 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
 
         if backprop and graphed_step is not None:
             if _is_fast_model(model_name):
-                # MMD randomness drawn OUTSIDE the hipGraph (static input)
+                # MMD randomness drawn OUTSIDE the hipGraph (static input);
+                # on the side stream so its allocations cannot alias
+                # graph-pool blocks (see GraphedStep.run_eager)
                 ns = train_config.mmd.samples * subgraphs
-                data.mmd_idx, data.mmd_valid = draw_sample_indices(
-                    data.batch, data.ptr, data.counts, ns)
+                data.mmd_idx, data.mmd_valid = graphed_step.run_eager(
+                    lambda: draw_sample_indices(data.batch, data.ptr,
+                                                data.counts, ns))
             (mse_log,) = graphed_step(data)
-            loss_accum = loss_accum + mse_log * batch_size
-            counter = counter + batch_size
+            # in-place: no eager allocation between replays
+            loss_accum.add_(mse_log, alpha=batch_size)
+            counter.add_(batch_size)
         else:
             total_node_cnt = data.counts_global.sum()
             node_cnt = float(data.num_nodes)
@@ -205,8 +209,8 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
             # node-count weighting: rank share of the global per-node MSE
             weight = node_cnt / total_node_cnt
             loss_loc = weight * loss_loc
-            loss_accum = loss_accum + loss_loc.detach() * batch_size
-            counter = counter + batch_size
+            loss_accum.add_(loss_loc.detach(), alpha=batch_size)
+            counter.add_(batch_size)
 
             if _is_fast_model(model_name) and virtual_node_loc is not None:
                 vloc = virtual_node_loc.permute(0, 2, 1).float()  # [B,C,3]
@@ -332,18 +336,24 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
             log_dict["loss_train"].append(loss_train)
 
         if epoch_index % log_config.test_interval == 0:
-            loss_valid = train_single_epoch(
-                rank, model, model_name, loader_valid, optimizer, scheduler,
-                loss_mse, config.data.dataset_name, train_config, epoch_index,
-                tag="valid", subgraphs=config.model.virtual_channels,
-                world_size=world_size, device=device,
-                autocast_dtype=autocast_dtype, progress=progress)
-            loss_test = train_single_epoch(
-                rank, model, model_name, loader_test, optimizer, scheduler,
-                loss_mse, config.data.dataset_name, train_config, epoch_index,
-                tag="test", subgraphs=config.model.virtual_channels,
-                world_size=world_size, device=device,
-                autocast_dtype=autocast_dtype, progress=progress)
+            def _eval(loader, tag):
+                return train_single_epoch(
+                    rank, model, model_name, loader, optimizer, scheduler,
+                    loss_mse, config.data.dataset_name, train_config,
+                    epoch_index, tag=tag,
+                    subgraphs=config.model.virtual_channels,
+                    world_size=world_size, device=device,
+                    autocast_dtype=autocast_dtype, progress=progress)
+
+            if graphed_step is not None:
+                # eval allocations must not alias captured-graph pools
+                loss_valid = graphed_step.run_eager(
+                    lambda: _eval(loader_valid, "valid"))
+                loss_test = graphed_step.run_eager(
+                    lambda: _eval(loader_test, "test"))
+            else:
+                loss_valid = _eval(loader_valid, "valid")
+                loss_test = _eval(loader_test, "test")
 
             if rank == 0:
                 log_dict["epochs"].append(epoch_index)
