@@ -320,18 +320,39 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
 
     early_stop_flag = torch.tensor(0, device=device)
 
+    # DISTEGNN_TORCH_PROFILE=<dir>: trace the FIRST epoch after warmup with
+    # torch.profiler (chrome trace per rank). rocprofv3 stays the primary
+    # kernel-level tool (profiles/); this covers host-side/op-level views.
+    profile_dir = os.environ.get("DISTEGNN_TORCH_PROFILE")
+
     for epoch_index in range(1 + start_epoch, train_config.epochs + 1):
         if early_stop_flag.item() == 1:
             print(f"Device {rank} stop succeed!")
             break
 
-        loss_train = train_single_epoch(
-            rank, model, model_name, loader_train, optimizer, scheduler,
-            loss_mse, config.data.dataset_name, train_config, epoch_index,
-            tag="train", subgraphs=config.model.virtual_channels,
-            world_size=world_size, device=device, grad_bucket=grad_bucket,
-            autocast_dtype=autocast_dtype, debug_lockstep=debug_lockstep,
-            progress=progress, graphed_step=graphed_step)
+        def _train_epoch():
+            return train_single_epoch(
+                rank, model, model_name, loader_train, optimizer, scheduler,
+                loss_mse, config.data.dataset_name, train_config,
+                epoch_index, tag="train",
+                subgraphs=config.model.virtual_channels,
+                world_size=world_size, device=device,
+                grad_bucket=grad_bucket, autocast_dtype=autocast_dtype,
+                debug_lockstep=debug_lockstep, progress=progress,
+                graphed_step=graphed_step)
+
+        if profile_dir and epoch_index == 2 + start_epoch:
+            from torch.profiler import (ProfilerActivity, profile)
+
+            with profile(activities=[ProfilerActivity.CPU,
+                                     ProfilerActivity.CUDA]) as prof:
+                loss_train = _train_epoch()
+            os.makedirs(profile_dir, exist_ok=True)
+            prof.export_chrome_trace(
+                os.path.join(profile_dir, f"trace_rank{rank}.json"))
+            profile_dir = None
+        else:
+            loss_train = _train_epoch()
         if rank == 0:
             log_dict["loss_train"].append(loss_train)
 

@@ -99,3 +99,26 @@ def test_exp_name_templates(tmp_path):
     cfg.data.inner_radius = 0.075
     name = entry.build_exp_name(cfg, 8)
     assert name.startswith("nbody_100_metis_FastEGNN_0.075_0.075_8_2_")
+
+
+def test_run_eager_cpu_passthrough():
+    from distegnn_amd.runtime.graphs import GraphedStep
+
+    g = GraphedStep(lambda b: (b,), [], enabled=True)  # no CUDA -> disabled
+    assert g.enabled is False
+    assert g.run_eager(lambda: 41 + 1) == 42
+
+
+def test_wandb_config_without_package(tmp_path, monkeypatch):
+    """wandb.enable=True must not crash when the package is missing."""
+    import yaml as _yaml
+
+    import main as entry
+
+    monkeypatch.setitem(sys.modules, "wandb", None)  # force ImportError
+    cfg = tiny_config(tmp_path)
+    cfg["log"]["wandb"]["enable"] = True
+    cfg_path = tmp_path / "cfg_wb.yaml"
+    cfg_path.write_text(_yaml.safe_dump(cfg))
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    entry.main(["--config_path", str(cfg_path)])
