@@ -124,7 +124,7 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
   }
 }
 
-__global__ __launch_bounds__(THREADS) void fused_edge_bwd(
+__global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
     const bf16* __restrict__ h, const float* __restrict__ coord,
     const float* __restrict__ eattr, const long* __restrict__ row,
     const long* __restrict__ col,
