@@ -101,11 +101,10 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
 
 def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
                world_size, device, clip=True, mmd_cfg=None):
-    data = batch.to(device)
-
     def _pre():
-        # eager work between replays runs on the side stream: its
-        # allocations must not alias captured-graph pool blocks
+        # eager work between replays (H2D copies included) runs on the side
+        # stream: its allocations must not alias captured-graph pool blocks
+        data = batch.to(device)
         data.counts_global = (comm.global_counts(data.counts)
                               if world_size > 1 else data.counts)
         if mmd_cfg is not None:
@@ -114,8 +113,9 @@ def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
 
             data.mmd_idx, data.mmd_valid = draw_sample_indices(
                 data.batch, data.ptr, data.counts, mmd_cfg)
+        return data
 
-    graphed.run_eager(_pre)
+    data = graphed.run_eager(_pre)
     (mse_log,) = graphed(data)
     if (step + 1) % accum == 0:
         if grad_bucket is not None:

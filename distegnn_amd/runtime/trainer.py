@@ -164,7 +164,12 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
         if step_timer is not None:
             step_timer.start()
         batch_size = data.num_graphs                # host-known, no sync
-        data = data.to(device)
+        if backprop and graphed_step is not None:
+            # H2D allocations must not come from default-stream blocks that
+            # alias captured-graph pools (GraphedStep.run_eager)
+            data = graphed_step.run_eager(lambda: data.to(device))
+        else:
+            data = data.to(device)
 
         if debug_lockstep and world_size > 1:
             gathered = [torch.zeros_like(data.loc_mean)
