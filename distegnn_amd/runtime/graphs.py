@@ -163,6 +163,19 @@ class GraphedStep:
             out_bufs = tuple(torch.empty(shape, dtype=dtype, device=dev)
                              for (shape, dtype, dev) in e.out_meta)
             torch.cuda.synchronize()
+            if (torch.distributed.is_available()
+                    and torch.distributed.is_initialized()):
+                # Drain the ProcessGroupNCCL watchdog before capturing:
+                # the watchdog thread polls hipEventQuery on outstanding
+                # pre-capture collectives (e.g. the per-step counts_global
+                # all-reduce) and an event query DURING stream capture is
+                # hipErrorStreamCaptureUnsupported -> process abort.
+                # Collectives issued INSIDE the capture are not watched.
+                import time as _time
+
+                torch.distributed.barrier()
+                torch.cuda.synchronize()
+                _time.sleep(0.5)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 outputs = self.step_fn(proxy)
