@@ -263,6 +263,10 @@ def main():
                                      op=torch.distributed.ReduceOp.MAX)
     ms_per_step = elapsed.item() * 1000.0 / args.steps
 
+    # destroy BEFORE printing: RCCL writes a version banner to stdout at
+    # teardown, and the driver expects the JSON line to come last
+    comm.barrier()
+    comm.destroy()
     if rank == 0:
         out = {
             "metric": "train_step_time_ms",
@@ -294,7 +298,6 @@ def main():
             },
         }
         print(json.dumps(out), flush=True)
-    comm.destroy()
 
 
 if __name__ == "__main__":
