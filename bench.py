@@ -118,10 +118,9 @@ def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
     data = graphed.run_eager(_pre)
     (mse_log,) = graphed(data)
     if (step + 1) % accum == 0:
-        if grad_bucket is not None:
-            grad_bucket.sync()
-
         def _opt():
+            if grad_bucket is not None:
+                grad_bucket.sync()     # flat-bucket cat allocates: side str.
             if clip:  # reference clip rule: FastEGNN + (ws>1 or LargeFluid)
                 torch.nn.utils.clip_grad_norm_(model_params(graphed),
                                                max_norm=0.3)

@@ -229,10 +229,9 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
 
         if backprop:
             if (step + 1) % train_config.accumulation_steps == 0:
-                if grad_bucket is not None:
-                    grad_bucket.sync()              # SUM == ref avg * ws
-
                 def _opt_region():
+                    if grad_bucket is not None:
+                        grad_bucket.sync()          # SUM == ref avg * ws
                     if ((world_size > 1 or dataset_name == "LargeFluid")
                             and model_name == "FastEGNN"):
                         nn.utils.clip_grad_norm_(model.parameters(),
