@@ -56,11 +56,37 @@ def init_distributed(backend: Optional[str] = None,
         )
     if backend == "nccl":
         torch.cuda.set_device(local_rank)
+        global _GRAPH_PG
+        if _GRAPH_PG is None:
+            # second communicator reserved for captured collectives (see
+            # _capture_group); warm it eagerly — NCCL comm init during a
+            # stream capture would be fatal
+            _GRAPH_PG = dist.new_group(backend="nccl")
+            warm = torch.ones(1, device=f"cuda:{local_rank}")
+            dist.all_reduce(warm, group=_GRAPH_PG)
+            torch.cuda.synchronize()
     return local_rank, world_size
+
+
+_GRAPH_PG = None  # dedicated communicator for hipGraph-captured collectives
 
 
 def is_distributed() -> bool:
     return dist.is_available() and dist.is_initialized()
+
+
+def _capture_group():
+    """The communicator for the CURRENT collective call.
+
+    NCCL/RCCL forbids mixing graph-captured and eager collectives on one
+    communicator: the eager per-step counts/gradient all-reduces were
+    observed to corrupt captured virtual-node all-reduces sharing the
+    default comm. Calls issued while a stream capture is active use a
+    dedicated process group (created and warmed eagerly at init)."""
+    if (_GRAPH_PG is not None and torch.cuda.is_available()
+            and torch.cuda.is_current_stream_capturing()):
+        return _GRAPH_PG
+    return None  # default group
 
 
 def world_size() -> int:
@@ -94,7 +120,7 @@ class _FusedAllReduceSum(torch.autograd.Function):
         if not is_distributed():
             return tensors if len(tensors) > 1 else tensors[0]
         flat = torch.cat([t.reshape(-1) for t in tensors])
-        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+        dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=_capture_group())
         outs = []
         ofs = 0
         for t in tensors:
@@ -108,7 +134,7 @@ class _FusedAllReduceSum(torch.autograd.Function):
         if not is_distributed():
             return grads if len(grads) > 1 else grads[0]
         flat = torch.cat([g.reshape(-1) for g in grads])
-        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+        dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=_capture_group())
         outs = []
         ofs = 0
         for g in grads:
