@@ -174,12 +174,13 @@ def main():
                     help="distinct synthetic samples to cycle through")
     ap.add_argument("--graphs", type=str, default="auto",
                     choices=["auto", "on", "off"],
-                    help="hipGraph-capture the fwd+loss+bwd step (auto = on; "
-                         "RCCL capture of the in-forward virtual-node "
-                         "all-reduce is validated on a 1-rank group via "
-                         "DISTEGNN_FORCE_DIST=1; asymmetric capture failure "
-                         "is deadlock-free because the eager fallback issues "
-                         "the same collectives in the same order)")
+                    help="hipGraph-capture the fwd+loss+bwd step; auto = on "
+                         "for 1 GPU, off for multi-GPU. Captured RCCL "
+                         "all-reduces replay with a corrupted logged loss "
+                         "on this RCCL build even on a dedicated capture "
+                         "communicator (reproduce: DISTEGNN_FORCE_DIST=1 "
+                         "torchrun -nproc 1 bench.py --graphs on), so the "
+                         "collective path defaults to eager execution")
     args = ap.parse_args()
 
     rank, world_size = comm.init_distributed()
@@ -235,7 +236,8 @@ def main():
 
     step_core = make_step_core(model, accum, mmd_sigma, mmd_samples,
                                autocast_dtype)
-    use_graphs = args.graphs != "off"
+    use_graphs = (args.graphs == "on"
+                  or (args.graphs == "auto" and ws_eff == 1))
     graphed = GraphedStep(step_core, model.parameters(),
                           warmup_occurrences=2,
                           enabled=use_graphs, verbose=True)
