@@ -144,6 +144,16 @@ class _FusedAllReduceSum(torch.autograd.Function):
         return tuple(outs)
 
 
+def _drain_watchdog():
+    """Let the ProcessGroupNCCL watchdog dequeue outstanding works before a
+    capture (its hipEventQuery during capture aborts the process)."""
+    import time as _time
+
+    dist.barrier()
+    torch.cuda.synchronize()
+    _time.sleep(0.5)
+
+
 def all_reduce_sum_differentiable(*tensors: torch.Tensor):
     """Fused differentiable all_reduce(SUM). Identity when world_size == 1."""
     return _FusedAllReduceSum.apply(*tensors)
@@ -220,6 +230,34 @@ class GradBucket:
             ofs += n
 
     @torch.no_grad()
+    def graph_sync(self):
+        """sync() as a captured graph replay (built lazily at first call,
+        which precedes the first step-graph capture for any
+        accumulation_steps <= warmup). Falls back to eager sync when not
+        distributed or not on CUDA."""
+        if not is_distributed():
+            return
+        if not torch.cuda.is_available():
+            return self.sync()
+        if getattr(self, "_sync_graph", None) is None:
+            for p in self.params:
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)
+            torch.cuda.synchronize()
+            _drain_watchdog()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                flat = torch.cat([p.grad.reshape(-1) for p in self.params])
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM,
+                                group=_capture_group())
+                ofs = 0
+                for p in self.params:
+                    n = p.numel()
+                    p.grad.copy_(flat[ofs:ofs + n].view_as(p.grad))
+                    ofs += n
+            self._sync_graph = g
+        self._sync_graph.replay()
+
     def sync(self):
         if not is_distributed():
             return
