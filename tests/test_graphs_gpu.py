@@ -114,3 +114,95 @@ def test_graph_replay_adam_mmd_long():
         assert b == b, (i, "nan under replay")
         assert b > -1e-6, (i, b, "negative mse under replay")
         assert abs(a - b) < 1e-4 + 0.3 * abs(a), (i, a, b)
+
+
+def test_graph_replay_with_captured_rccl_collectives(tmp_path):
+    """DistEGNN path under capture on a 1-rank RCCL group: the in-forward
+    virtual-node all-reduces and the flat gradient sync run as captured
+    graphs. Guards the three bisected replay-window hazards (eager
+    default-stream allocs, watchdog polls during capture, device-wide
+    syncs between replays) end-to-end."""
+    import torch.distributed as dist
+
+    from distegnn_amd.parallel import comm
+    from distegnn_amd.parallel.comm import GradBucket
+    from distegnn_amd.runtime.losses import draw_sample_indices, mmd_loss
+
+    import distegnn_amd.parallel.comm as C
+
+    if not dist.is_initialized():
+        dist.init_process_group(
+            "nccl", rank=0, world_size=1,
+            init_method=f"file://{tmp_path}/pg_init")
+    # graph-capture communicator (normally made by comm.init_distributed)
+    if C._GRAPH_PG is None:
+        C._GRAPH_PG = dist.new_group(backend="nccl")
+        warm = torch.ones(1, device="cuda:0")
+        dist.all_reduce(warm, group=C._GRAPH_PG)
+        torch.cuda.synchronize()
+
+    def run(enabled):
+        model, batches = build(42)
+        # world_size=2 flips the model's distributed branch; the 1-rank
+        # group makes every collective an identity
+        model.world_size = 2
+        for layer in model.modules():
+            if hasattr(layer, "world_size"):
+                layer.world_size = 2
+        gb = GradBucket(model)
+        opt = torch.optim.Adam(model.parameters(), lr=5e-4)
+
+        def fn(data):
+            cg = comm.global_counts(data.counts)
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                loc, vloc = model(data.x, data.pos, data.vel,
+                                  data.loc_mean, data.edge_index,
+                                  data.batch, edge_attr=data.edge_attr,
+                                  rowptr=data.rowptr, ptr=data.ptr,
+                                  counts=data.counts, counts_global=cg,
+                                  colptr=data.colptr,
+                                  col_perm=data.col_perm)
+            loss = torch.nn.functional.mse_loss(loc.float(), data.target)
+            mse = loss.detach()
+            lm = mmd_loss(vloc.permute(0, 2, 1).float(), data.target,
+                          data.batch, data.ptr, data.counts, 1.5, 3,
+                          sample_idx=data.mmd_idx,
+                          sample_valid=data.mmd_valid)
+            (loss + 0.01 * lm).backward()
+            return (mse,)
+
+        g = GraphedStep(fn, model.parameters(), warmup_occurrences=2,
+                        enabled=enabled)
+        losses = []
+        for k in range(20):
+            data = batches[k % 2]
+
+            def _pre():
+                data.mmd_idx, data.mmd_valid = draw_sample_indices(
+                    data.batch, data.ptr, data.counts, 9)
+
+            g.run_eager(_pre)
+            (loss,) = g(data)
+
+            def _opt():
+                if g.enabled:
+                    gb.graph_sync()
+                else:
+                    gb.sync()
+                opt.step()
+                opt.zero_grad(set_to_none=False)
+
+            g.run_eager(_opt)
+            losses.append(loss.item())
+        return losses
+
+    try:
+        eager = run(False)
+        graphed = run(True)
+        for i, (a, b) in enumerate(zip(eager, graphed)):
+            assert b == b and b > -1e-6, (i, b)
+            assert abs(a - b) < 1e-4 + 0.3 * abs(a), (i, a, b)
+    finally:
+        torch.cuda.synchronize()
+        C._GRAPH_PG = None
+        dist.destroy_process_group()
