@@ -83,3 +83,32 @@ def test_synthetic_fields_fluid():
     s = make_cloud_sample("Fluid113K", rng, n_override=1000)
     assert s["x"].shape == (1000, 3)      # [visc, mass, |v|]
     assert s["attr"].shape == (1000, 2)   # [visc, mass]
+
+
+def test_nbody_generator_smoke(tmp_path):
+    """The offline N-body generator produces reader-compatible .npy files."""
+    import os
+    import subprocess
+    import sys
+
+    import numpy as np
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = os.path.join(root, "dataset_generation", "nbody",
+                          "generate_dataset.py")
+    out = tmp_path / "nb"
+    subprocess.run(
+        [sys.executable, script, "--path", str(out), "--num-train", "2",
+         "--num-valid", "1", "--num-test", "1", "--length", "200",
+         "--sample-freq", "50", "--n_isolated", "12", "--clusters", "2",
+         "--seed", "7"],
+        check=True, timeout=300)
+    files = sorted(p.name for p in out.iterdir())
+    assert any(f.startswith("loc_train") for f in files), files
+    loc = np.load(next(out.glob("loc_train*.npy")))
+    vel = np.load(next(out.glob("vel_train*.npy")))
+    q = np.load(next(out.glob("charges_train*.npy")))
+    assert loc.shape[0] == 2 and loc.shape[2] == 12 and loc.shape[-1] == 3
+    assert vel.shape == loc.shape
+    assert q.shape[0] == 2 and q.shape[1] == 12
+    assert np.isfinite(loc).all() and np.isfinite(vel).all()
