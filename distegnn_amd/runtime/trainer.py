@@ -278,7 +278,12 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                           os.environ.get("DISTEGNN_DEBUG_LOCKSTEP") == "1")
 
     # hipGraph-captured train step (config train.hip_graphs: auto|on|off;
-    # auto = on for single-GPU CUDA runs)
+    # auto = on for single-GPU CUDA runs). Multi-GPU capture stays off by
+    # default HERE (bench.py carries the full captured-collective design):
+    # the epoch loop's logging/early-stop all-reduces would run as eager
+    # RCCL between replays, which corrupts captured state on this stack —
+    # they would need captured-graph variants first (see
+    # parallel/comm.py GradBucket.graph_sync and profiles/README.md).
     hg = str(train_config.get("hip_graphs", "auto")).lower()
     use_graphs = (hg in ("on", "true") or
                   (hg == "auto" and world_size == 1)) \
