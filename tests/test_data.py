@@ -112,3 +112,88 @@ def test_nbody_generator_smoke(tmp_path):
     assert vel.shape == loc.shape
     assert q.shape[0] == 2 and q.shape[1] == 12
     assert np.isfinite(loc).all() and np.isfinite(vel).all()
+
+
+def test_batch_csr_invariants_fuzz():
+    """Property test of the CSR contract every HIP kernel relies on:
+    row-sorted edges, exact rowptr/colptr, col_perm sorting, chunk-table
+    coverage — across random graph counts/sizes/densities."""
+    from distegnn_amd.data.graph import Data, collate
+
+    g = torch.Generator().manual_seed(123)
+    for trial in range(20):
+        n_graphs = int(torch.randint(1, 6, (1,), generator=g))
+        datas = []
+        for _ in range(n_graphs):
+            n = int(torch.randint(2, 90, (1,), generator=g))
+            m = int(torch.randint(0, 4 * n, (1,), generator=g))
+            ei = torch.randint(0, n, (2, m), generator=g)
+            keep = ei[0] != ei[1]
+            ei = ei[:, keep]
+            m = ei.size(1)
+            datas.append(Data(x=torch.rand(n, 2, generator=g),
+                              pos=torch.rand(n, 3, generator=g),
+                              vel=torch.rand(n, 3, generator=g),
+                              target=torch.rand(n, 3, generator=g),
+                              edge_index=ei,
+                              edge_attr=torch.rand(m, 2, generator=g),
+                              loc_mean=torch.rand(1, 3, generator=g)))
+        b = collate(datas)
+        n, m = b.num_nodes, b.num_edges
+        row, col = b.edge_index[0], b.edge_index[1]
+        # row-sorted
+        assert (row[1:] >= row[:-1]).all()
+        # rowptr is the exact CSR of row
+        assert b.rowptr.numel() == n + 1
+        assert b.rowptr[0] == 0 and b.rowptr[-1] == m
+        for seg in range(n):
+            s, e = int(b.rowptr[seg]), int(b.rowptr[seg + 1])
+            assert (row[s:e] == seg).all()
+        # col_perm sorts col; colptr is the CSR of the sorted col
+        sc = col[b.col_perm]
+        assert (sc[1:] >= sc[:-1]).all()
+        assert b.colptr[0] == 0 and b.colptr[-1] == m
+        for seg in range(n):
+            s, e = int(b.colptr[seg]), int(b.colptr[seg + 1])
+            assert (sc[s:e] == seg).all()
+        # edges stay within their graph block
+        for gi in range(b.num_graphs):
+            s, e = int(b.ptr[gi]), int(b.ptr[gi + 1])
+            mask = (row >= s) & (row < e)
+            assert ((col[mask] >= s) & (col[mask] < e)).all()
+        # chunk tables (when built) tile [ptr[i], ptr[i+1]) exactly
+        if b.pool_chunk_begin is not None:
+            cb, ce = b.pool_chunk_begin, b.pool_chunk_end
+            scp = b.pool_seg_chunk_ptr
+            assert scp[0] == 0 and scp[-1] == cb.numel()
+            for gi in range(b.num_graphs):
+                ks = int(scp[gi])
+                ke = int(scp[gi + 1])
+                assert int(cb[ks]) == int(b.ptr[gi])
+                assert int(ce[ke - 1]) == int(b.ptr[gi + 1])
+                for k in range(ks + 1, ke):
+                    assert int(cb[k]) == int(ce[k - 1])
+
+
+def test_batch_chunk_tables_large_segment():
+    """A >4096-node graph triggers the chunk tables; validate coverage."""
+    from distegnn_amd.data.graph import Data, collate
+
+    g = torch.Generator().manual_seed(5)
+    n = 5000
+    ei = torch.randint(0, n, (2, 100), generator=g)
+    ei = ei[:, ei[0] != ei[1]]
+    b = collate([Data(x=torch.rand(n, 2, generator=g),
+                      pos=torch.rand(n, 3, generator=g),
+                      vel=torch.rand(n, 3, generator=g),
+                      target=torch.rand(n, 3, generator=g),
+                      edge_index=ei,
+                      edge_attr=torch.rand(ei.size(1), 2, generator=g),
+                      loc_mean=torch.rand(1, 3, generator=g))])
+    assert b.pool_chunk_begin is not None
+    cb, ce, scp = (b.pool_chunk_begin, b.pool_chunk_end,
+                   b.pool_seg_chunk_ptr)
+    assert scp.tolist() == [0, cb.numel()]
+    assert int(cb[0]) == 0 and int(ce[-1]) == n
+    assert (cb[1:] == ce[:-1]).all()
+    assert ((ce - cb) <= 256).all() and ((ce - cb) > 0).all()
