@@ -244,7 +244,8 @@ def main():
     use_graphs = args.graphs != "off"
     graphed = GraphedStep(step_core, model.parameters(),
                           warmup_occurrences=2,
-                          enabled=use_graphs, verbose=True)
+                          enabled=use_graphs, verbose=True,
+                          fallback_ctx=comm.capture_comm_fallback)
 
     clip = args.workload == "largefluid" or world_size > 1
     num_sample = mmd_samples * vch

@@ -75,18 +75,41 @@ def is_distributed() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
+_FORCE_CAPTURE_COMM = False
+
+
 def _capture_group():
     """The communicator for the CURRENT collective call.
 
     NCCL/RCCL forbids mixing graph-captured and eager collectives on one
-    communicator: the eager per-step counts/gradient all-reduces were
-    observed to corrupt captured virtual-node all-reduces sharing the
-    default comm. Calls issued while a stream capture is active use a
-    dedicated process group (created and warmed eagerly at init)."""
-    if (_GRAPH_PG is not None and torch.cuda.is_available()
-            and torch.cuda.is_current_stream_capturing()):
+    communicator. Calls issued while a stream capture is active use a
+    dedicated process group (created and warmed eagerly at init); the
+    same group is used under ``capture_comm_fallback`` so that a rank
+    whose capture failed keeps issuing the SAME collective sequence on
+    the SAME communicator as the ranks that replay captured graphs
+    (asymmetric fallback must not become a comm-mismatch deadlock)."""
+    if _GRAPH_PG is not None and (
+            _FORCE_CAPTURE_COMM
+            or (torch.cuda.is_available()
+                and torch.cuda.is_current_stream_capturing())):
         return _GRAPH_PG
     return None  # default group
+
+
+class capture_comm_fallback:
+    """Context manager: route collectives to the capture communicator
+    even though no capture is active (eager fallback of a graphed step)."""
+
+    def __enter__(self):
+        global _FORCE_CAPTURE_COMM
+        self._prev = _FORCE_CAPTURE_COMM
+        _FORCE_CAPTURE_COMM = True
+        return self
+
+    def __exit__(self, *exc):
+        global _FORCE_CAPTURE_COMM
+        _FORCE_CAPTURE_COMM = self._prev
+        return False
 
 
 def world_size() -> int:

@@ -70,7 +70,8 @@ class GraphedStep:
     """
 
     def __init__(self, step_fn, params, warmup_occurrences: int = 2,
-                 enabled: bool = True, verbose: bool = False):
+                 enabled: bool = True, verbose: bool = False,
+                 fallback_ctx=None):
         self.step_fn = step_fn
         self.params = list(params)
         self.warmup = warmup_occurrences
@@ -78,6 +79,11 @@ class GraphedStep:
         self.verbose = verbose
         self.entries: Dict[Tuple, _ShapeEntry] = {}
         self._side = None  # side stream for eager work between replays
+        # context manager factory wrapped around POST-capture eager
+        # fallbacks (a rank whose capture failed must keep issuing its
+        # collectives on the capture communicator — comm mismatch with
+        # replaying ranks would deadlock)
+        self.fallback_ctx = fallback_ctx
 
     def run_eager(self, fn):
         """Run eager work (optimizer step, clip, zero_grad) between replays.
@@ -102,6 +108,12 @@ class GraphedStep:
         torch.cuda.current_stream().wait_stream(self._side)
         return out
 
+    def _fallback(self, batch):
+        if self.fallback_ctx is None:
+            return self.step_fn(batch)
+        with self.fallback_ctx():
+            return self.step_fn(batch)
+
     @staticmethod
     def _key(batch) -> Tuple:
         return (batch.num_nodes, batch.num_edges, batch.num_graphs)
@@ -124,7 +136,7 @@ class GraphedStep:
         key = self._key(batch)
         e = self.entries.setdefault(key, _ShapeEntry())
         if e.disabled:
-            return self.step_fn(batch)
+            return self._fallback(batch)
         if e.graph is not None:
             self._copy_into(e.static, batch)
             e.graph.replay()
@@ -197,4 +209,4 @@ class GraphedStep:
             if self.verbose:
                 print(f"[graphs] capture failed for {key}: {exc}; eager")
             torch.cuda.synchronize()
-            return self.step_fn(batch)
+            return self._fallback(batch)
