@@ -1,3 +1,7 @@
+"""1-rank forced-dist repro harness for the RCCL replay-window hazards:
+--barrier injects an eager dist.barrier() between replays (corrupts),
+--midsync injects a bare torch.cuda.synchronize() (corrupts),
+no flags = the clean captured-collective path. See profiles/README.md."""
 import os, sys
 sys.path.insert(0, "/root/repo")
 os.environ.setdefault("DISTEGNN_FORCE_DIST", "1")

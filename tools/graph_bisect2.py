@@ -1,3 +1,8 @@
+"""Bisection harness for the hipGraph training corruption (water3d):
+compares eager vs graphed loss trajectories with switchable MMD variants
+(--mmd-idx static indices, --no-mmd, --manual-cdist, --no-gather,
+--detach-mmd, --one-batch, --probe). Kept as evidence/repro for the
+side-stream allocator hazard documented in runtime/graphs.py."""
 import sys, torch
 sys.path.insert(0, "/root/repo")
 import bench
