@@ -50,7 +50,7 @@ def make_loaders(dataset_train, dataset_valid, dataset_test, batch_size: int,
         return _TorchDataLoader(
             ds, batch_size=batch_size, sampler=sampler, shuffle=shuffle,
             drop_last=True, num_workers=num_workers, collate_fn=collate,
-            pin_memory=False,
+            pin_memory=torch.cuda.is_available(),
         )
 
     return (
