@@ -22,6 +22,7 @@ from __future__ import annotations
 
 import os
 import random
+import zlib
 from typing import List
 
 import numpy as np
@@ -91,7 +92,7 @@ def _process_nbody_cutoff(data_config) -> List[str]:
         if _is_synthetic(data_config):
             data = synth.make_cutoff_dataset(
                 "nbody_100", _num_samples(data_config, partition),
-                seed=hash(partition) % (2 ** 31),
+                seed=zlib.crc32(partition.encode()) % (2 ** 31),
                 radius=data_config.radius,
                 cutoff_rate=data_config.cutoff_rate)
             torch.save(data, path)
@@ -141,7 +142,7 @@ def _process_cutoff_generic(data_config) -> List[str]:
                 f"to use the synthetic generator at published scales")
         data = synth.make_cutoff_dataset(
             data_config.dataset_name, _num_samples(data_config, partition),
-            seed=hash(partition) % (2 ** 31), radius=data_config.radius,
+            seed=zlib.crc32(partition.encode()) % (2 ** 31), radius=data_config.radius,
             cutoff_rate=data_config.cutoff_rate)
         torch.save(data, path)
         print(f"{path} processed!")
@@ -186,7 +187,7 @@ def process_dataset_distribute(rank: int, world_size: int, data_config
         per_rank = synth.make_distributed_dataset(
             name, _num_samples(data_config, partition), world_size,
             split_mode=data_config.split_mode,
-            seed=hash(partition) % (2 ** 31),
+            seed=zlib.crc32(partition.encode()) % (2 ** 31),
             outer_radius=data_config.outer_radius,
             inner_radius=data_config.inner_radius,
             n_override=data_config.get("synthetic_nodes", None))

@@ -122,3 +122,35 @@ def test_wandb_config_without_package(tmp_path, monkeypatch):
     cfg_path.write_text(_yaml.safe_dump(cfg))
     monkeypatch.setenv("WORLD_SIZE", "1")
     entry.main(["--config_path", str(cfg_path)])
+
+
+def test_golden_loss_trajectory(tmp_path, monkeypatch):
+    """Deterministic CPU golden run: guards the end-to-end math (model,
+    loss weighting, MMD, optimizer wiring) against silent regressions.
+    Record mode: DISTEGNN_UPDATE_GOLDEN=1 rewrites the stored values."""
+    import numpy as np
+
+    import main as entry
+
+    cfg = tiny_config(tmp_path)
+    cfg["train"]["epochs"] = 3
+    cfg["log"]["test_interval"] = 3
+    cfg_path = tmp_path / "cfg_gold.yaml"
+    cfg_path.write_text(yaml.safe_dump(cfg))
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    entry.main(["--config_path", str(cfg_path)])
+
+    logs = tmp_path / "logs"
+    exp = next(logs.iterdir())
+    hist = json.loads((exp / "log" / "log.json").read_text())[1]
+    got = hist["loss_train"]
+
+    golden_path = os.path.join(os.path.dirname(__file__),
+                               "golden_nbody_losses.json")
+    if os.environ.get("DISTEGNN_UPDATE_GOLDEN") == "1":
+        with open(golden_path, "w") as f:
+            json.dump(got, f)
+    golden = json.load(open(golden_path))
+    assert len(got) == len(golden)
+    for a, b in zip(got, golden):
+        assert abs(a - b) < 1e-6 + 1e-4 * abs(b), (got, golden)
