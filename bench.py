@@ -216,7 +216,8 @@ def main():
                                      nodes, radius, args.split_mode,
                                      seed=43)
     else:
-        assert world_size == 1,             "cutoff-mode workloads are single-device (reference main.py:173)"
+        assert world_size == 1, \
+            "cutoff-mode workloads are single-device (reference main.py:173)"
         batches = build_cutoff_batches(args.workload, args.num_batches,
                                        graphs_per_batch, seed=43)
     if rank == 0:
