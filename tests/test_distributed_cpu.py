@@ -307,3 +307,42 @@ def test_main_two_rank_distribute_mode(tmp_path):
              args=(str(init_file), str(cfg_path),
                    str(tmp_path / "logs")),
              nprocs=2, join=True)
+
+
+def test_capture_comm_fallback_flag():
+    """The fallback context toggles the module flag and restores on exit
+    (used so post-capture eager fallbacks keep the capture communicator)."""
+    import distegnn_amd.parallel.comm as C
+
+    assert C._FORCE_CAPTURE_COMM is False
+    with C.capture_comm_fallback():
+        assert C._FORCE_CAPTURE_COMM is True
+        with C.capture_comm_fallback():
+            assert C._FORCE_CAPTURE_COMM is True
+        assert C._FORCE_CAPTURE_COMM is True
+    assert C._FORCE_CAPTURE_COMM is False
+    # off-GPU there is never a capture group
+    assert C._capture_group() is None
+
+
+def test_graphedstep_fallback_ctx_used():
+    """GraphedStep routes disabled-shape fallbacks through fallback_ctx."""
+    from distegnn_amd.runtime.graphs import GraphedStep
+
+    calls = []
+
+    class Ctx:
+        def __enter__(self):
+            calls.append("enter")
+
+        def __exit__(self, *a):
+            calls.append("exit")
+            return False
+
+    g = GraphedStep(lambda b: ("out",), [], enabled=True, fallback_ctx=Ctx)
+    assert g.enabled is False  # no CUDA here -> plain path, ctx unused
+    assert g(object()) == ("out",)
+    assert calls == []
+    # the internal fallback path itself must wrap with the ctx
+    assert g._fallback(object()) == ("out",)
+    assert calls == ["enter", "exit"]
