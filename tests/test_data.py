@@ -197,3 +197,23 @@ def test_batch_chunk_tables_large_segment():
     assert int(cb[0]) == 0 and int(ce[-1]) == n
     assert (cb[1:] == ce[:-1]).all()
     assert ((ce - cb) <= 256).all() and ((ce - cb) > 0).all()
+
+
+def test_cutoff_edge_drops_longest():
+    """cutoff_edge keeps the shortest (1-rate) fraction (reference
+    process_dataset.py:300-305)."""
+    from distegnn_amd.data.preprocess import cutoff_edge
+
+    g = torch.Generator().manual_seed(3)
+    pos = torch.rand(50, 3, generator=g)
+    ei = torch.randint(0, 50, (2, 200), generator=g)
+    ei = ei[:, ei[0] != ei[1]]
+    m = ei.size(1)
+    out = cutoff_edge(ei, pos, 0.25)
+    assert out.size(1) == int(m * 0.75)
+    d_all = (pos[ei[0]] - pos[ei[1]]).norm(dim=1)
+    d_kept = (pos[out[0]] - pos[out[1]]).norm(dim=1)
+    thresh = torch.sort(d_all).values[int(m * 0.75) - 1]
+    assert (d_kept <= thresh + 1e-6).all()
+    # rate 0 is the identity
+    assert torch.equal(cutoff_edge(ei, pos, 0.0), ei)
