@@ -346,3 +346,68 @@ def test_graphedstep_fallback_ctx_used():
     # the internal fallback path itself must wrap with the ctx
     assert g._fallback(object()) == ("out",)
     assert calls == ["enter", "exit"]
+
+
+def _rf_worker(rank, init_file, result_dir):
+    from distegnn_amd.models.fastrf import FastRF
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=2)
+    try:
+        per_rank, _ = _make_partitions()
+        fix_seed(11)
+        model = FastRF(edge_attr_nf=2, hidden_nf=32, virtual_channels=3,
+                       world_size=2, n_layers=2).double()
+        batch = collate(per_rank[rank])
+        counts = batch.counts.double()
+        cg = counts.clone()
+        dist.all_reduce(cg)
+        loc, vloc = model(batch.pos.double(), batch.vel.double(),
+                          batch.loc_mean.double(), batch.edge_index,
+                          batch.batch, edge_attr=batch.edge_attr.double(),
+                          rowptr=batch.rowptr, ptr=batch.ptr, counts=counts,
+                          counts_global=cg)
+        loc.pow(2).sum().backward()
+        grads = {n: (p.grad.clone() if p.grad is not None
+                     else torch.zeros_like(p))
+                 for n, p in model.named_parameters()}
+        torch.save({"loc": loc.detach(), "vloc": vloc.detach(),
+                    "grads": grads},
+                   os.path.join(result_dir, f"rf{rank}.pt"))
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_fastrf_equals_merged_graph(tmp_path):
+    """FastRF's single virtual-exchange site: 2-rank partitioned forward
+    equals the merged-graph forward; summed local grads equal merged grads."""
+    from distegnn_amd.models.fastrf import FastRF
+
+    init_file = tmp_path / "pg_init_rf"
+    mp.spawn(_rf_worker, args=(str(init_file), str(tmp_path)), nprocs=2,
+             join=True)
+    r0 = torch.load(tmp_path / "rf0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "rf1.pt", weights_only=False)
+
+    per_rank, merged = _make_partitions()
+    fix_seed(11)
+    model = FastRF(edge_attr_nf=2, hidden_nf=32, virtual_channels=3,
+                   world_size=1, n_layers=2).double()
+    mb = collate(merged)
+    loc, vloc = model(mb.pos.double(), mb.vel.double(),
+                      mb.loc_mean.double(), mb.edge_index, mb.batch,
+                      edge_attr=mb.edge_attr.double(), rowptr=mb.rowptr,
+                      ptr=mb.ptr, counts=mb.counts.double())
+    loc.pow(2).sum().backward()
+
+    got = torch.cat([r0["loc"], r1["loc"]])
+    # fp64 sums cross the weighted-average reduce in a different order
+    # than the merged-graph pooling -> 1e-7 instead of exact
+    assert torch.allclose(got, loc.detach(), atol=1e-7)
+    assert torch.allclose(r0["vloc"], vloc.detach(), atol=1e-7)
+    for n, p in model.named_parameters():
+        g = p.grad if p.grad is not None else torch.zeros_like(p)
+        assert torch.allclose(r0["grads"][n] + r1["grads"][n], g,
+                              atol=1e-6, rtol=1e-6), n
