@@ -154,3 +154,18 @@ def test_golden_loss_trajectory(tmp_path, monkeypatch):
     assert len(got) == len(golden)
     for a, b in zip(got, golden):
         assert abs(a - b) < 1e-6 + 1e-4 * abs(b), (got, golden)
+
+
+def test_torch_profile_hook(tmp_path, monkeypatch):
+    """DISTEGNN_TORCH_PROFILE=<dir> dumps a chrome trace of one epoch."""
+    import main as entry
+
+    cfg = tiny_config(tmp_path)
+    cfg["train"]["epochs"] = 2
+    cfg_path = tmp_path / "cfg_prof.yaml"
+    cfg_path.write_text(yaml.safe_dump(cfg))
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    monkeypatch.setenv("DISTEGNN_TORCH_PROFILE", str(tmp_path / "prof"))
+    entry.main(["--config_path", str(cfg_path)])
+    trace = tmp_path / "prof" / "trace_rank0.json"
+    assert trace.exists() and trace.stat().st_size > 1000
