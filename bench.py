@@ -69,6 +69,11 @@ def make_step_core(model, accum, mmd_sigma, mmd_samples, autocast_dtype):
     reduced eagerly by the caller before the graphed region)."""
 
     def step_core(data):
+        # captured collectives (this counts reduce + the in-forward virtual
+        # exchanges) route to the dedicated capture communicator via
+        # comm._capture_group(); eager collectives (teardown barrier, timing
+        # MAX-reduce) stay on the default group — NCCL forbids mixing
+        # captured and eager work on one communicator
         counts_global = comm.global_counts(data.counts)
         total_node_cnt = counts_global.sum()
         chunks = None
@@ -176,6 +181,13 @@ def main():
     ap.add_argument("--virtual-channels", type=int, default=None)
     ap.add_argument("--num-batches", type=int, default=2,
                     help="distinct synthetic samples to cycle through")
+    ap.add_argument("--integrity", type=str, default="auto",
+                    choices=["auto", "on", "off"],
+                    help="self-validate capture/replay before the measured "
+                         "phase: run a short trajectory eager and graphed "
+                         "from identical state, compare losses, and disable "
+                         "capture coherently on all ranks on divergence "
+                         "(auto = on for multi-rank graphed runs)")
     ap.add_argument("--graphs", type=str, default="auto",
                     choices=["auto", "on", "off"],
                     help="hipGraph-capture the fwd+loss+bwd step (auto = "
@@ -250,6 +262,28 @@ def main():
 
     clip = args.workload == "largefluid" or world_size > 1
     num_sample = mmd_samples * vch
+
+    if ws_eff > 1 and grad_bucket is not None and use_graphs:
+        # build the captured grad-sync graph BEFORE any step capture (its
+        # watchdog drain is a default-group barrier — must not land inside
+        # the replay window)
+        grad_bucket.prebuild_graph_sync()
+
+    run_gate = (args.integrity == "on"
+                or (args.integrity == "auto" and use_graphs and ws_eff > 1))
+    if run_gate and graphed.enabled:
+        from distegnn_amd.runtime.integrity import run_capture_integrity_gate
+
+        def _gate_step(k):
+            return train_step(graphed, batches[k % len(batches)], optimizer,
+                              grad_bucket, k, accum, ws_eff, device, clip,
+                              mmd_cfg=num_sample)
+
+        n_check = 4 * len(batches)   # warmup_occurrences(2) + 2 replays/key
+        run_capture_integrity_gate(graphed, _gate_step, n_check,
+                                   list(model.parameters()), optimizer,
+                                   rank=rank)
+
     mse = None
     for w in range(args.warmup):
         mse = train_step(graphed, batches[w % len(batches)], optimizer,
@@ -262,7 +296,9 @@ def main():
     # call; tools/fd_debug.py --barrier / --midsync). Ranks stay in
     # lockstep through the captured per-step collectives; timing uses
     # device events and the MAX-reduce below absorbs residual skew.
-    event_timing = use_graphs and ws_eff > 1
+    # graphed.enabled (not use_graphs): the integrity gate may have disabled
+    # capture coherently — the eager path then uses barriers safely
+    event_timing = graphed.enabled and ws_eff > 1
     if event_timing:
         ev0 = torch.cuda.Event(enable_timing=True)
         ev1 = torch.cuda.Event(enable_timing=True)
@@ -326,6 +362,7 @@ def main():
                 "virtual_channels": vch,
                 "accumulation_steps": accum,
                 "parallelism": f"graph-partition dp{world_size}",
+                "hip_graphs": bool(graphed.enabled),
                 "coord_mse": final_mse,
             },
         }

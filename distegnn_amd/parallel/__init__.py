@@ -1,7 +1,9 @@
 from .comm import (
+    CapturedAllReduce,
     GradBucket,
     all_reduce_sum_differentiable,
     barrier,
+    capture_comm_fallback,
     check_model_parameters,
     destroy,
     fused_weighted_average_reduce,
@@ -13,8 +15,8 @@ from .comm import (
 )
 
 __all__ = [
-    "GradBucket", "all_reduce_sum_differentiable", "barrier",
-    "check_model_parameters", "destroy", "fused_weighted_average_reduce",
-    "global_counts", "init_distributed", "is_distributed", "rank",
-    "world_size",
+    "CapturedAllReduce", "GradBucket", "all_reduce_sum_differentiable",
+    "barrier", "capture_comm_fallback", "check_model_parameters", "destroy",
+    "fused_weighted_average_reduce", "global_counts", "init_distributed",
+    "is_distributed", "rank", "world_size",
 ]

@@ -187,12 +187,79 @@ def global_counts(counts: torch.Tensor) -> torch.Tensor:
 
     The reference recomputes these weights with a host-synced python loop at
     every reduce site (FastEGNN.py:196,226,260); they are constant within a
-    step, so one tiny collective replaces 12+ host syncs + 24 reduces."""
+    step, so one tiny collective replaces 12+ host syncs + 24 reduces.
+
+    Routed through ``_capture_group()``: when issued inside a hipGraph
+    capture (bench/trainer step_core) or under ``capture_comm_fallback``,
+    the reduce lands on the dedicated capture communicator — mixing it onto
+    the default group would violate NCCL's captured/eager single-communicator
+    rule (the teardown barrier and timing reduces run eagerly on the default
+    group)."""
     if not is_distributed():
         return counts
     out = counts.clone()
-    dist.all_reduce(out, op=dist.ReduceOp.SUM)
+    dist.all_reduce(out, op=dist.ReduceOp.SUM, group=_capture_group())
     return out
+
+
+class CapturedAllReduce:
+    """All-reduce as a captured-graph replay, for collectives that must run
+    INSIDE the replay window of an RCCL-containing step graph (per-epoch
+    logging reduce, early-stop flag — profiles/README.md roadmap #2).
+
+    An eager collective on the DEFAULT communicator between replays of a
+    captured RCCL graph corrupts replayed outputs on this stack (bisected:
+    tools/fd_debug.py --barrier). Replaying a pre-captured graph on the
+    capture communicator keeps every post-capture collective on one
+    communicator with a rank-consistent schedule.
+
+    ``prebuild`` MUST be called before the first step-graph capture: building
+    lazily would interleave the build's watchdog drain (a default-group
+    barrier) into the replay window. Non-CUDA / non-distributed calls fall
+    back to an eager all_reduce (gloo path, CPU tests)."""
+
+    def __init__(self):
+        self._entries = {}
+
+    @staticmethod
+    def _key(t: torch.Tensor, op) -> tuple:
+        return (tuple(t.shape), t.dtype, str(op))
+
+    def _build(self, t: torch.Tensor, op):
+        buf = t.clone()
+        _drain_watchdog()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            dist.all_reduce(buf, op=op, group=_capture_group())
+        self._entries[self._key(t, op)] = (g, buf)
+
+    def prebuild(self, tensors_ops):
+        """tensors_ops: iterable of (example_tensor, ReduceOp). Call during
+        setup, before any step-graph capture."""
+        if not is_distributed() or not torch.cuda.is_available():
+            return
+        for t, op in tensors_ops:
+            if t.is_cuda and self._key(t, op) not in self._entries:
+                self._build(t, op)
+
+    def __call__(self, t: torch.Tensor, op=None) -> torch.Tensor:
+        op = op if op is not None else dist.ReduceOp.SUM
+        if not is_distributed():
+            return t
+        if not (torch.cuda.is_available() and t.is_cuda):
+            out = t.clone()
+            dist.all_reduce(out, op=op)
+            return out
+        entry = self._entries.get(self._key(t, op))
+        if entry is None:  # late build: caller did not prebuild
+            self._build(t, op)
+            entry = self._entries[self._key(t, op)]
+        g, buf = entry
+        buf.copy_(t)
+        g.replay()
+        # the STATIC buffer is returned (no allocation inside the replay
+        # window): read/copy it before the next call with the same key
+        return buf
 
 
 def fused_weighted_average_reduce(tensors: Sequence[torch.Tensor],
@@ -252,12 +319,18 @@ class GradBucket:
             p.data.copy_(flat[ofs:ofs + n].view_as(p))
             ofs += n
 
+    def prebuild_graph_sync(self):
+        """Build the captured grad-sync graph during setup, BEFORE the first
+        step-graph capture — a lazy build's watchdog drain (default-group
+        barrier) must not land inside the replay window."""
+        if is_distributed() and torch.cuda.is_available():
+            self.graph_sync(_build_only=True)
+
     @torch.no_grad()
-    def graph_sync(self):
-        """sync() as a captured graph replay (built lazily at first call,
-        which precedes the first step-graph capture for any
-        accumulation_steps <= warmup). Falls back to eager sync when not
-        distributed or not on CUDA."""
+    def graph_sync(self, _build_only: bool = False):
+        """sync() as a captured graph replay (prebuild at setup via
+        ``prebuild_graph_sync``; lazily built otherwise). Falls back to
+        eager sync when not distributed or not on CUDA."""
         if not is_distributed():
             return
         if not torch.cuda.is_available():
@@ -279,7 +352,8 @@ class GradBucket:
                     p.grad.copy_(flat[ofs:ofs + n].view_as(p.grad))
                     ofs += n
             self._sync_graph = g
-        self._sync_graph.replay()
+        if not _build_only:
+            self._sync_graph.replay()
 
     def sync(self):
         if not is_distributed():

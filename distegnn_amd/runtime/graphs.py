@@ -71,7 +71,7 @@ class GraphedStep:
 
     def __init__(self, step_fn, params, warmup_occurrences: int = 2,
                  enabled: bool = True, verbose: bool = False,
-                 fallback_ctx=None):
+                 fallback_ctx=None, max_entries: int = 32):
         self.step_fn = step_fn
         self.params = list(params)
         self.warmup = warmup_occurrences
@@ -79,11 +79,19 @@ class GraphedStep:
         self.verbose = verbose
         self.entries: Dict[Tuple, _ShapeEntry] = {}
         self._side = None  # side stream for eager work between replays
-        # context manager factory wrapped around POST-capture eager
-        # fallbacks (a rank whose capture failed must keep issuing its
-        # collectives on the capture communicator — comm mismatch with
-        # replaying ranks would deadlock)
+        # context manager factory wrapped around EVERY eager invocation
+        # while capture is enabled (warmup occurrences, post-capture
+        # fallbacks, overflow keys): per-rank shape keys mean one rank can
+        # be replaying (collectives on the capture communicator) while
+        # another runs the same step eagerly — routing the eager rank's
+        # collectives to the same communicator keeps the per-communicator
+        # schedule rank-consistent (comm mismatch would deadlock)
         self.fallback_ctx = fallback_ctx
+        # graph pools cannot be freed safely; real datasets have per-sample
+        # shapes, so cap captured entries — overflow keys run eagerly (with
+        # fallback_ctx, so the collective schedule stays consistent)
+        self.max_entries = max_entries
+        self._overflow_warned = False
 
     def run_eager(self, fn):
         """Run eager work (optimizer step, clip, zero_grad) between replays.
@@ -116,7 +124,16 @@ class GraphedStep:
 
     @staticmethod
     def _key(batch) -> Tuple:
-        return (batch.num_nodes, batch.num_edges, batch.num_graphs)
+        # (num_nodes, num_edges, num_graphs) alone does not distinguish
+        # batches whose optional static fields differ in shape or presence
+        # (e.g. pool-chunk tables at equal totals) — _copy_into would then
+        # copy_ with mismatched shapes at replay. Key on the full field
+        # signature instead.
+        sig = tuple(
+            (f, tuple(v.shape), v.dtype)
+            for f in _BATCH_FIELDS
+            if torch.is_tensor(v := getattr(batch, f, None)))
+        return (batch.num_nodes, batch.num_edges, batch.num_graphs, sig)
 
     def _snapshot(self, batch) -> dict:
         out = {}
@@ -134,7 +151,17 @@ class GraphedStep:
         if not self.enabled:
             return self.step_fn(batch)
         key = self._key(batch)
-        e = self.entries.setdefault(key, _ShapeEntry())
+        e = self.entries.get(key)
+        if e is None:
+            captured = sum(1 for v in self.entries.values()
+                           if v.graph is not None or not v.disabled)
+            if captured >= self.max_entries:
+                if not self._overflow_warned:
+                    self._overflow_warned = True
+                    print(f"[graphs] shape-key cache full "
+                          f"({self.max_entries}); further shapes run eager")
+                return self._fallback(batch)
+            e = self.entries.setdefault(key, _ShapeEntry())
         if e.disabled:
             return self._fallback(batch)
         if e.graph is not None:
@@ -143,7 +170,9 @@ class GraphedStep:
             return e.outputs
         e.seen += 1
         if e.seen <= self.warmup or e.out_meta is None:
-            outs = self.step_fn(batch)
+            # warmup occurrences run eagerly; collectives still go to the
+            # capture communicator (another rank may already be replaying)
+            outs = self._fallback(batch)
             e.out_meta = [(tuple(o.shape), o.dtype, o.device)
                           for o in outs]
             return outs

@@ -60,14 +60,17 @@ def load_state_dict_compat(model: nn.Module, sd: dict):
     model.load_state_dict(sd)
 
 
-def model_forward(model: nn.Module, model_name: str, data, device):
+def model_forward(model: nn.Module, model_name: str, data, device,
+                  counts_global=None):
     """Per-model forward dispatch (reference utils/train.py:63-90)."""
     chunks = None
     if getattr(data, "pool_chunk_begin", None) is not None:
         chunks = (data.pool_chunk_begin, data.pool_chunk_end,
                   data.pool_seg_chunk_ptr)
+    if counts_global is None:
+        counts_global = getattr(data, "counts_global", None)
     kw = dict(rowptr=data.rowptr, ptr=data.ptr, counts=data.counts,
-              counts_global=getattr(data, "counts_global", None),
+              counts_global=counts_global,
               pool_chunks=chunks, colptr=getattr(data, "colptr", None),
               col_perm=getattr(data, "col_perm", None))
     if model_name in ("FastEGNN", "FastSchNet"):
@@ -106,20 +109,28 @@ def _is_fast_model(model_name: str) -> bool:
 
 
 def make_train_step_core(model, model_name, loss_fn, train_config,
-                         autocast_dtype, device):
+                         autocast_dtype, device, world_size=1):
     """forward + loss (+MMD) + accumulation-scaled backward on
-    device-resident batch tensors — hipGraph-capturable (no host syncs,
-    no collectives; counts_global is reduced eagerly by the caller)."""
+    device-resident batch tensors — hipGraph-capturable (no host syncs;
+    the counts reduce and the in-forward virtual exchanges are captured
+    collectives on the dedicated capture communicator, see
+    parallel/comm.py)."""
 
     def step_core(data):
-        total_node_cnt = data.counts_global.sum()
+        # in-graph counts reduce from the static counts input (capture-
+        # communicator routed). Always recomputed — NEVER read from a
+        # pre-set batch attribute, which would bake one batch's reduced
+        # counts into the captured graph as a stale static
+        counts_global = (comm.global_counts(data.counts)
+                         if world_size > 1 else data.counts)
+        total_node_cnt = counts_global.sum()
         node_cnt = float(data.num_nodes)
         with contextlib.ExitStack() as stack:
             if autocast_dtype is not None:
                 stack.enter_context(torch.autocast("cuda",
                                                    dtype=autocast_dtype))
             loc_pred, virtual_node_loc = model_forward(
-                model, model_name, data, device)
+                model, model_name, data, device, counts_global=counts_global)
         loss_loc = loss_fn(loc_pred.float(), data.target)
         weight = node_cnt / total_node_cnt
         loss_loc = weight * loss_loc
@@ -142,7 +153,8 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
                        loss_fn, dataset_name, train_config, epoch_index, tag,
                        subgraphs, world_size, device, grad_bucket=None,
                        autocast_dtype=None, debug_lockstep=False,
-                       progress=True, step_timer=None, graphed_step=None):
+                       progress=True, step_timer=None, graphed_step=None,
+                       epoch_reduce=None):
     backprop = tag == "train"
     if backprop:
         model.train()
@@ -171,7 +183,11 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
         else:
             data = data.to(device)
 
-        if debug_lockstep and world_size > 1:
+        graphed_train = backprop and graphed_step is not None
+        if debug_lockstep and world_size > 1 and not graphed_train:
+            # an eager default-group all_gather inside the replay window of
+            # a captured-RCCL step corrupts replayed outputs (bisected) —
+            # the lockstep debug check is only available with hip_graphs off
             gathered = [torch.zeros_like(data.loc_mean)
                         for _ in range(world_size)]
             dist.all_gather(gathered, data.loc_mean)
@@ -181,9 +197,11 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
                                           atol=1e-6), \
                         "train loader out of lockstep across ranks"
 
-        # one collective per step: per-graph global node counts
-        data.counts_global = comm.global_counts(data.counts) \
-            if world_size > 1 else data.counts
+        if not graphed_train:
+            # one collective per step: per-graph global node counts (the
+            # graphed path reduces counts INSIDE the captured step instead)
+            data.counts_global = comm.global_counts(data.counts) \
+                if world_size > 1 else data.counts
 
         if backprop and graphed_step is not None:
             if _is_fast_model(model_name):
@@ -231,7 +249,12 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
             if (step + 1) % train_config.accumulation_steps == 0:
                 def _opt_region():
                     if grad_bucket is not None:
-                        grad_bucket.sync()          # SUM == ref avg * ws
+                        if graphed_step is not None and graphed_step.enabled:
+                            # captured replay: eager RCCL between replays
+                            # corrupts captured state on this stack
+                            grad_bucket.graph_sync()
+                        else:
+                            grad_bucket.sync()      # SUM == ref avg * ws
                     if ((world_size > 1 or dataset_name == "LargeFluid")
                             and model_name == "FastEGNN"):
                         nn.utils.clip_grad_norm_(model.parameters(),
@@ -250,15 +273,25 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
         if step_timer is not None:
             step_timer.stop()
 
-    # deferred logging reduce: one collective per epoch (ref: per step)
+    # deferred logging reduce: one collective per epoch (ref: per step).
+    # Under a graphed multi-GPU run this must be a captured-graph replay on
+    # the capture communicator (epoch_reduce = CapturedAllReduce) — an eager
+    # default-group collective between replays is a bisected corruption.
     if world_size > 1:
-        dist.all_reduce(loss_accum, op=dist.ReduceOp.SUM)
+        if epoch_reduce is not None:
+            loss_accum = epoch_reduce(loss_accum, dist.ReduceOp.SUM)
+        else:
+            dist.all_reduce(loss_accum, op=dist.ReduceOp.SUM)
     if counter.item() == 0:
         if rank == 0:
             print(f"WARNING: {tag} loader produced no batches (dataset "
                   f"smaller than batch_size with drop_last) — skipping")
         return float("nan")
-    avg = (loss_accum / counter).item()
+    if graphed_step is not None:
+        # scalar math between replays allocates on the side stream
+        avg = graphed_step.run_eager(lambda: (loss_accum / counter).item())
+    else:
+        avg = (loss_accum / counter).item()
     if rank == 0:
         prefix = "" if backprop else "==> "
         print(f"{prefix}{tag} epoch: {epoch_index}, avg loss: {avg:.5f}")
@@ -278,22 +311,42 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                           os.environ.get("DISTEGNN_DEBUG_LOCKSTEP") == "1")
 
     # hipGraph-captured train step (config train.hip_graphs: auto|on|off;
-    # auto = on for single-GPU CUDA runs). Multi-GPU capture stays off by
-    # default HERE (bench.py carries the full captured-collective design):
-    # the epoch loop's logging/early-stop all-reduces would run as eager
-    # RCCL between replays, which corrupts captured state on this stack —
-    # they would need captured-graph variants first (see
-    # parallel/comm.py GradBucket.graph_sync and profiles/README.md).
+    # auto = on for every CUDA run, single- AND multi-GPU). The multi-GPU
+    # captured-collective design (profiles/README.md roadmap #2): the
+    # in-forward virtual exchanges and the counts reduce are captured
+    # INSIDE the step graph on the dedicated capture communicator, the
+    # gradient sync is its own captured graph (GradBucket.graph_sync), the
+    # per-epoch logging/early-stop reduces replay pre-captured graphs
+    # (CapturedAllReduce), eval runs eagerly with its collectives routed to
+    # the capture communicator, and no default-group collective or
+    # device-wide sync happens inside the replay window. A pre-flight
+    # integrity gate (train.graph_integrity_check, default on for ws>1)
+    # compares a short eager vs replayed trajectory and disables capture
+    # coherently on all ranks on divergence.
     hg = str(train_config.get("hip_graphs", "auto")).lower()
-    use_graphs = (hg in ("on", "true") or
-                  (hg == "auto" and world_size == 1)) \
-        and device.type == "cuda"
+    use_graphs = hg in ("on", "true", "auto") and device.type == "cuda"
     graphed_step = None
+    epoch_reduce = None
+    eval_ctx = contextlib.nullcontext
     if use_graphs:
         graphed_step = GraphedStep(
             make_train_step_core(model, model_name, loss_mse, train_config,
-                                 autocast_dtype, device),
-            model.parameters(), warmup_occurrences=2)
+                                 autocast_dtype, device,
+                                 world_size=world_size),
+            model.parameters(), warmup_occurrences=2,
+            fallback_ctx=(comm.capture_comm_fallback if world_size > 1
+                          else None))
+        if world_size > 1:
+            eval_ctx = comm.capture_comm_fallback
+            if grad_bucket is not None:
+                grad_bucket.prebuild_graph_sync()
+            captured_reduce = comm.CapturedAllReduce()
+            captured_reduce.prebuild([
+                (torch.zeros((), device=device), dist.ReduceOp.SUM),
+                (torch.zeros((), dtype=torch.int64, device=device),
+                 dist.ReduceOp.MAX),
+            ])
+            epoch_reduce = captured_reduce
 
     log_dict = {"epochs": [], "loss": [], "loss_train": []}
     best_log_dict = {"epoch_index": 0, "loss_valid": 1e8, "loss_test": 1e8,
@@ -329,6 +382,58 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
 
     early_stop_flag = torch.tensor(0, device=device)
 
+    # Pre-flight capture integrity gate (default on for graphed multi-GPU
+    # runs): train a short trajectory twice from identical state — eager,
+    # then captured/replayed — and disable capture coherently on all ranks
+    # if the losses diverge. See runtime/integrity.py.
+    gic = str(train_config.get("graph_integrity_check", "auto")).lower()
+    run_gate = graphed_step is not None and graphed_step.enabled and (
+        gic in ("on", "true") or (gic == "auto" and world_size > 1))
+    if run_gate:
+        import itertools
+
+        from .integrity import run_capture_integrity_gate
+
+        gate_batches = list(itertools.islice(iter(loader_train), 2))
+        if gate_batches:
+            model.train()
+            optimizer.zero_grad(set_to_none=False)
+            accum = train_config.accumulation_steps
+
+            def _gate_step(k):
+                data = gate_batches[k % len(gate_batches)]
+                data = graphed_step.run_eager(lambda: data.to(device))
+                if _is_fast_model(model_name):
+                    ns = (train_config.mmd.samples
+                          * config.model.virtual_channels)
+                    data.mmd_idx, data.mmd_valid = graphed_step.run_eager(
+                        lambda: draw_sample_indices(data.batch, data.ptr,
+                                                    data.counts, ns))
+                (mse_log,) = graphed_step(data)
+                if (k + 1) % accum == 0:
+                    def _opt():
+                        if grad_bucket is not None:
+                            if graphed_step.enabled:
+                                grad_bucket.graph_sync()
+                            else:
+                                grad_bucket.sync()
+                        if ((world_size > 1 or
+                             config.data.dataset_name == "LargeFluid")
+                                and model_name == "FastEGNN"):
+                            nn.utils.clip_grad_norm_(model.parameters(),
+                                                     max_norm=0.3)
+                        optimizer.step()
+                        optimizer.zero_grad(set_to_none=False)
+
+                    graphed_step.run_eager(_opt)
+                return mse_log
+
+            run_capture_integrity_gate(
+                graphed_step, _gate_step,
+                n_steps=(graphed_step.warmup + 2) * len(gate_batches),
+                params=list(model.parameters()), optimizer=optimizer,
+                rank=rank)
+
     # DISTEGNN_TORCH_PROFILE=<dir>: trace the FIRST epoch after warmup with
     # torch.profiler (chrome trace per rank). rocprofv3 stays the primary
     # kernel-level tool (profiles/); this covers host-side/op-level views.
@@ -348,7 +453,7 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                 world_size=world_size, device=device,
                 grad_bucket=grad_bucket, autocast_dtype=autocast_dtype,
                 debug_lockstep=debug_lockstep, progress=progress,
-                graphed_step=graphed_step)
+                graphed_step=graphed_step, epoch_reduce=epoch_reduce)
 
         if profile_dir and epoch_index == 2 + start_epoch:
             from torch.profiler import (ProfilerActivity, profile)
@@ -366,14 +471,26 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
             log_dict["loss_train"].append(loss_train)
 
         if epoch_index % log_config.test_interval == 0:
+            # the reference evaluates in fp32 (no autocast in
+            # utils/train.py's no-grad epochs): eval/checkpoint selection
+            # stays fp32 unless train.bf16_eval is set explicitly
+            eval_dtype = (autocast_dtype
+                          if train_config.get("bf16_eval", False) else None)
+
             def _eval(loader, tag):
-                return train_single_epoch(
-                    rank, model, model_name, loader, optimizer, scheduler,
-                    loss_mse, config.data.dataset_name, train_config,
-                    epoch_index, tag=tag,
-                    subgraphs=config.model.virtual_channels,
-                    world_size=world_size, device=device,
-                    autocast_dtype=autocast_dtype, progress=progress)
+                # under a graphed multi-GPU run, eval's eager collectives
+                # (counts reduce, in-forward virtual exchanges) are routed
+                # to the capture communicator (eval_ctx) — the default
+                # group must stay quiet inside the replay window
+                with eval_ctx():
+                    return train_single_epoch(
+                        rank, model, model_name, loader, optimizer,
+                        scheduler, loss_mse, config.data.dataset_name,
+                        train_config, epoch_index, tag=tag,
+                        subgraphs=config.model.virtual_channels,
+                        world_size=world_size, device=device,
+                        autocast_dtype=eval_dtype, progress=progress,
+                        epoch_reduce=epoch_reduce)
 
             if graphed_step is not None:
                 # eval allocations must not alias captured-graph pools
@@ -421,7 +538,11 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                 print(f"Early stopped! Epoch: {epoch_index}")
                 early_stop_flag.fill_(1)
             if world_size > 1:
-                dist.all_reduce(early_stop_flag, op=dist.ReduceOp.MAX)
+                if epoch_reduce is not None:
+                    early_stop_flag.copy_(
+                        epoch_reduce(early_stop_flag, dist.ReduceOp.MAX))
+                else:
+                    dist.all_reduce(early_stop_flag, op=dist.ReduceOp.MAX)
 
         if rank == 0:
             best_log_dict["time_cost"] = time.perf_counter() - start
