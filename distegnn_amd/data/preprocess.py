@@ -10,12 +10,14 @@ Dispatchers (reference datasets/process_dataset.py:32-58):
   rank-i transport; the cache key bakes in the partition count
   (``..._{ws}_{outer:.3f}_{inner:.3f}_{max_samples}_{dt}__{rank}-{ws}.pt``).
 
-Real-data readers are implemented where this image has the codec (nbody
-.npy via numpy); Water-3D needs h5py and Fluid113K needs zstandard — absent
-here, those readers raise with a clear message unless
-``data_config.synthetic`` is set, in which case samples come from
-data/synthetic.py at the same scales and are cached under the same key
-scheme with a ``synthetic-`` dataset prefix.
+Real-data readers (data/readers/) cover all four workloads: nbody .npy via
+numpy, Water-3D HDF5 via h5py-or-hdf5lite, Fluid113K msgpack(.zst) with a
+built-in msgpack_numpy-compatible codec (zstd decompression needs the
+zstandard package; uncompressed .msgpack chunks are accepted), protein via
+guarded MDAnalysis with the test-split rot/trans augmentation
+(reference :162-174) shared with the synthetic path.
+``data_config.synthetic`` switches to data/synthetic.py generators at the
+same scales, cached under the same key scheme with a ``synthetic-`` prefix.
 """
 
 from __future__ import annotations
@@ -60,10 +62,12 @@ def cutoff_edge(edge_index: torch.Tensor, pos: torch.Tensor,
 
 def process_dataset_edge_cutoff(data_config) -> List[str]:
     name = data_config.dataset_name
-    if name == "nbody_100":
+    if name.startswith("nbody"):
         return _process_nbody_cutoff(data_config)
-    if name in ("protein", "Water-3D"):
-        return _process_cutoff_generic(data_config)
+    if name.startswith("protein"):
+        return _process_protein_cutoff(data_config)
+    if name.startswith("Water-3D"):
+        return _process_water3d_cutoff(data_config)
     raise NotImplementedError(f"cutoff mode for dataset {name}")
 
 
@@ -124,26 +128,99 @@ def _process_nbody_cutoff(data_config) -> List[str]:
     return paths
 
 
-def _process_cutoff_generic(data_config) -> List[str]:
-    """Water-3D / protein cutoff-mode datasets.
+def _process_water3d_cutoff(data_config) -> List[str]:
+    """Water-3D cutoff mode (reference :225-297): per-trajectory HDF5 groups,
+    15 random frames per trajectory up to max_samples. Real files are read
+    via h5py when installed, else the in-tree hdf5lite subset reader."""
+    from .readers.common import build_cutoff_sample
+    from .readers.water3d import iter_water3d_trajectories, sample_frames
 
-    Real readers need h5py (Water-3D, reference :225-297) or MDAnalysis
-    (protein, reference :128-222) — not in this image; synthetic mode
-    reproduces the scales."""
     paths = _cutoff_cache_names(data_config, f"{data_config.delta_t}")
+    dt = data_config.delta_t
     for partition, path in zip(("train", "valid", "test"), paths):
         if os.path.exists(path):
             print(f"{path} exists!")
             continue
-        if not _is_synthetic(data_config):
-            raise NotImplementedError(
-                f"real-data reader for {data_config.dataset_name} requires "
-                f"h5py/MDAnalysis (absent in this image); set data.synthetic "
-                f"to use the synthetic generator at published scales")
-        data = synth.make_cutoff_dataset(
-            data_config.dataset_name, _num_samples(data_config, partition),
-            seed=zlib.crc32(partition.encode()) % (2 ** 31), radius=data_config.radius,
-            cutoff_rate=data_config.cutoff_rate)
+        if _is_synthetic(data_config):
+            data = synth.make_cutoff_dataset(
+                data_config.dataset_name,
+                _num_samples(data_config, partition),
+                seed=zlib.crc32(partition.encode()) % (2 ** 31),
+                radius=data_config.radius,
+                cutoff_rate=data_config.cutoff_rate)
+            torch.save(data, path)
+            continue
+        file_path = os.path.join(data_config.data_dir,
+                                 data_config.dataset_name,
+                                 f"{partition}.h5")
+        if not os.path.exists(file_path):
+            raise FileNotFoundError(
+                f"{file_path} not found: place the Water-3D HDF5 files "
+                f"(dataset_generation/Water-3D/tfrecord_to_h5.py) under "
+                f"data_dir, or set data.synthetic")
+        data = []
+        for _key, ptype, pos in iter_water3d_trajectories(file_path):
+            budget = data_config.max_samples - len(data)
+            if budget <= 0:
+                break
+            # reference draws from [0, 250] (:249); clamp for short
+            # trajectories so frame+delta_t stays in range
+            frames = sample_frames(15, budget,
+                                   max_frame=min(250, pos.size(0) - dt - 2))
+            for frame in frames:
+                data.append(build_cutoff_sample(
+                    pos[frame], pos[frame + 1] - pos[frame],
+                    pos[frame + dt], ptype, data_config.radius,
+                    data_config.cutoff_rate))
+        torch.save(data, path)
+        print(f"{path} processed!")
+    return paths
+
+
+def _process_protein_cutoff(data_config) -> List[str]:
+    """Protein (AdK) cutoff mode (reference :128-222) with the test-split
+    rot/trans augmentation (:162-174) honored for BOTH the real MDAnalysis
+    reader and the synthetic generator (config protein_fastegnn.yaml:20-21
+    test_rot/test_trans)."""
+    from .readers.common import apply_test_augmentation, build_cutoff_sample
+    from .readers.protein import (TRAIN_VALID_TEST_SPLIT, load_adk,
+                                  read_frame_triplet)
+
+    paths = _cutoff_cache_names(data_config, f"{data_config.delta_t}")
+    test_rot = bool(data_config.get("test_rot", False))
+    test_trans = bool(data_config.get("test_trans", False))
+    for partition, path in zip(("train", "valid", "test"), paths):
+        if os.path.exists(path):
+            print(f"{path} exists!")
+            continue
+        data = []
+        if _is_synthetic(data_config):
+            rng = torch.Generator().manual_seed(
+                zlib.crc32(partition.encode()) % (2 ** 31))
+            for _ in range(_num_samples(data_config, partition)):
+                s = synth.make_cloud_sample("protein", rng)
+                loc_0, vel_0, loc_t = s["pos"], s["vel"], s["target"]
+                if partition == "test":
+                    loc_0, vel_0, loc_t = apply_test_augmentation(
+                        loc_0, vel_0, loc_t, test_rot, test_trans,
+                        box=np.full(3, float(loc_0.max() - loc_0.min())))
+                data.append(build_cutoff_sample(
+                    loc_0, vel_0, loc_t, s["attr"], data_config.radius,
+                    data_config.cutoff_rate))
+        else:
+            universe, atom_ix, charges, n_frames, _dims = load_adk(
+                data_config.data_dir,
+                backbone=bool(data_config.get("backbone", True)))
+            lo, hi = TRAIN_VALID_TEST_SPLIT[partition]
+            for t in range(lo, hi):
+                loc_0, vel_0, loc_t, box = read_frame_triplet(
+                    universe, atom_ix, t, data_config.delta_t)
+                if partition == "test":
+                    loc_0, vel_0, loc_t = apply_test_augmentation(
+                        loc_0, vel_0, loc_t, test_rot, test_trans, box=box)
+                data.append(build_cutoff_sample(
+                    loc_0, vel_0, loc_t, charges, data_config.radius,
+                    data_config.cutoff_rate))
         torch.save(data, path)
         print(f"{path} processed!")
     return paths
@@ -178,22 +255,108 @@ def process_dataset_distribute(rank: int, world_size: int, data_config
         if all(os.path.exists(fname(partition, r)) for r in range(world_size)):
             print(f"{fname(partition, 0)} (and peers) exist!")
             continue
-        if not _is_synthetic(data_config):
-            raise NotImplementedError(
-                f"real-data reader for {name} requires "
-                f"{'zstandard' if name == 'Fluid113K' else 'h5py'} (absent "
-                f"in this image); set data.synthetic for synthetic data at "
-                f"published scales")
-        per_rank = synth.make_distributed_dataset(
-            name, _num_samples(data_config, partition), world_size,
-            split_mode=data_config.split_mode,
-            seed=zlib.crc32(partition.encode()) % (2 ** 31),
-            outer_radius=data_config.outer_radius,
-            inner_radius=data_config.inner_radius,
-            n_override=data_config.get("synthetic_nodes", None))
+        if _is_synthetic(data_config):
+            per_rank = synth.make_distributed_dataset(
+                name, _num_samples(data_config, partition), world_size,
+                split_mode=data_config.split_mode,
+                seed=zlib.crc32(partition.encode()) % (2 ** 31),
+                outer_radius=data_config.outer_radius,
+                inner_radius=data_config.inner_radius,
+                n_override=data_config.get("synthetic_nodes", None))
+        elif name == "Fluid113K":
+            per_rank = _read_fluid_dist_partition(data_config, partition,
+                                                  world_size)
+        else:
+            per_rank = _read_water3d_dist_partition(data_config, partition,
+                                                    world_size)
         for i in range(world_size - 1):
             assert len(per_rank[i]) == len(per_rank[i + 1])
         for i in range(world_size):
             torch.save(per_rank[i], fname(partition, i))
             print(f"{fname(partition, i)} processed!")
     return mine
+
+
+def _split_sample(data_config, world_size, pos, x, target, vel, attr):
+    """Dispatch one frame to the configured graph splitter (reference
+    :348-396 / :507-555)."""
+    mode = data_config.split_mode
+    if mode == "random":
+        return SPLITTERS["random"](
+            pos=pos, x=x, target=target, vel=vel, attr=attr,
+            radius=data_config.inner_radius, world_size=world_size,
+            device="cpu")
+    return SPLITTERS[mode](
+        pos=pos, x=x, target=target, vel=vel, attr=attr,
+        outer_radius=data_config.outer_radius,
+        inner_radius=data_config.inner_radius, world_size=world_size,
+        device="cpu")
+
+
+def _read_water3d_dist_partition(data_config, partition: str,
+                                 world_size: int) -> List[List[Data]]:
+    """Water-3D distribute mode, real HDF5 reader (reference :308-438)."""
+    from .readers.water3d import iter_water3d_trajectories, sample_frames
+
+    file_path = os.path.join(data_config.data_dir,
+                             data_config.dataset_name, f"{partition}.h5")
+    if not os.path.exists(file_path):
+        raise FileNotFoundError(
+            f"{file_path} not found: place the Water-3D HDF5 files under "
+            f"data_dir, or set data.synthetic")
+    dt = data_config.delta_t
+    per_rank: List[List[Data]] = [[] for _ in range(world_size)]
+    for _key, ptype, pos in iter_water3d_trajectories(file_path):
+        budget = data_config.max_samples - len(per_rank[0])
+        if budget <= 0:
+            break
+        frames = sample_frames(15, budget,
+                               max_frame=min(250, pos.size(0) - dt - 2))
+        for frame in frames:
+            vel_frame = pos[frame + 1] - pos[frame]
+            node_feat = torch.cat(
+                [vel_frame.pow(2).sum(-1, keepdim=True).sqrt(),
+                 ptype / ptype.max()], dim=-1)
+            parts = _split_sample(data_config, world_size, pos[frame],
+                                  node_feat, pos[frame + dt], vel_frame,
+                                  ptype)
+            for i, p in enumerate(parts):
+                per_rank[i].append(p)
+    return per_rank
+
+
+# reference :441-446: Fluid113K sims 1-101 / 101-121 / 121-141
+_FLUID_SIM_SPLIT = {"train": (1, 101), "valid": (101, 121),
+                    "test": (121, 141)}
+
+
+def _read_fluid_dist_partition(data_config, partition: str,
+                               world_size: int) -> List[List[Data]]:
+    """Fluid113K distribute mode, real msgpack(.zst) reader (reference
+    :441-578): 16-chunk sims, 16 random frames per sim from [0, 50],
+    node_attr = [viscosity, mass], node_feat = [attr, |v|]."""
+    from .readers.fluid113k import read_fluid_sim
+
+    base = os.path.join(data_config.data_dir, data_config.dataset_name)
+    dt = data_config.delta_t
+    per_rank: List[List[Data]] = [[] for _ in range(world_size)]
+    lo, hi = _FLUID_SIM_SPLIT[partition]
+    for idx in range(lo, hi):
+        budget = data_config.max_samples - len(per_rank[0])
+        if budget <= 0:
+            break
+        position, vel, viscosity, mass = read_fluid_sim(base, idx)
+        max_frame = min(50, position.size(0) - dt - 1)
+        frames = [random.randint(0, max_frame)
+                  for _ in range(min(16, budget))]
+        node_attr = torch.stack([viscosity, mass], dim=-1)
+        for frame in frames:
+            node_feat = torch.cat(
+                [node_attr,
+                 vel[frame].pow(2).sum(-1, keepdim=True).sqrt()], dim=-1)
+            parts = _split_sample(data_config, world_size, position[frame],
+                                  node_feat, position[frame + dt],
+                                  vel[frame], node_attr)
+            for i, p in enumerate(parts):
+                per_rank[i].append(p)
+    return per_rank

@@ -126,8 +126,10 @@ def barrier():
 
 
 def destroy():
+    global _GRAPH_PG
     if is_distributed():
         dist.destroy_process_group()
+    _GRAPH_PG = None
 
 
 class _FusedAllReduceSum(torch.autograd.Function):
