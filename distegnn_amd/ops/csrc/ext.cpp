@@ -42,6 +42,12 @@ std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
     torch::Tensor w3v, bool normalize, double eps);
+std::vector<torch::Tensor> fused_edge_backward_wg(
+    torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
+    torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
+    torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    torch::Tensor w3v, bool normalize, double eps);
 
 torch::Tensor radius_graph(torch::Tensor pos, double r) {
   return std::get<0>(radius_graph_gpu(pos, r));
@@ -81,6 +87,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("normalize"), py::arg("eps"));
   m.def("fused_edge_backward", &fused_edge_backward,
         "fused edge-block backward: in-LDS recompute + per-edge grads",
+        py::arg("h"), py::arg("coord"), py::arg("eattr"), py::arg("row"),
+        py::arg("col"), py::arg("dmsg_n"), py::arg("dtrans_n"),
+        py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
+        py::arg("w3"), py::arg("b3"), py::arg("w3v"), py::arg("normalize"),
+        py::arg("eps"));
+  m.def("fused_edge_backward_wg", &fused_edge_backward_wg,
+        "edge-block backward with in-kernel MFMA weight gradients "
+        "(no per-edge intermediates): {dhr, dhc, dcd, dw3v, gb, gw1, "
+        "gw2, gw3}",
         py::arg("h"), py::arg("coord"), py::arg("eattr"), py::arg("row"),
         py::arg("col"), py::arg("dmsg_n"), py::arg("dtrans_n"),
         py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),

@@ -105,6 +105,53 @@ __device__ __forceinline__ bf16x8 lds8_silu(const char* smem, int off) {
   return r;
 }
 
+// Transposed fragment: A[i][k] = T[k][i] for a row-major LDS tile
+// T[edge][c] with byte stride `stride` — 8 strided u16 reads per lane.
+// Used by the in-kernel wgrad GEMMs, whose contraction runs over the
+// EDGE dimension (the tile row index).
+__device__ __forceinline__ bf16x8 lds8_t(const char* smem, int base,
+                                         int stride, int i, int k0) {
+  bf16x8 v;
+#pragma unroll
+  for (int u = 0; u < 8; ++u)
+    v[u] = *reinterpret_cast<const __bf16*>(smem + base +
+                                            (k0 + u) * stride + i * 2);
+  return v;
+}
+__device__ __forceinline__ bf16x8 lds8_t_silu(const char* smem, int base,
+                                              int stride, int i, int k0) {
+  bf16x8 v = lds8_t(smem, base, stride, i, k0);
+#pragma unroll
+  for (int u = 0; u < 8; ++u) v[u] = (__bf16)silu_((float)v[u]);
+  return v;
+}
+
+// In-kernel weight-gradient accumulation: dW[i][j] += sum_e dz[e][i] *
+// act[e][j] over this tile's 64 edges (2 MFMA k-steps). Each wave owns
+// i-tile `wave`; NJ j-tiles; padded edges contribute zero because their
+// dz rows are exactly zero.
+template <int NJ, bool SILU_B>
+__device__ __forceinline__ void wg_acc(const char* smem, int a_base,
+                                       int a_stride, int b_base,
+                                       int b_stride, int wave, int lane,
+                                       f32x4 (&acc)[NJ]) {
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    int k0 = kk * 32 + (lane >> 4) * 8;
+    bf16x8 a = lds8_t(smem, a_base, a_stride, wave * 16 + (lane & 15), k0);
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+      bf16x8 b = SILU_B
+                     ? lds8_t_silu(smem, b_base, b_stride,
+                                   j * 16 + (lane & 15), k0)
+                     : lds8_t(smem, b_base, b_stride, j * 16 + (lane & 15),
+                              k0);
+      acc[j] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[j], 0, 0, 0);
+    }
+  }
+}
+
 // A from LDS (optionally through silu), B from global [64][wk] k-contig.
 template <int KSTEPS, bool SILU_A>
 __device__ __forceinline__ void mm_g(const char* smem, int a_off,
@@ -124,7 +171,15 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
   }
 }
 
-__global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
+// FUSE_WG: accumulate the three weight gradients (dW1 = dz1^T ein,
+// dW2 = dz2^T silu(z1), dW3 = dz3^T silu(z2)) in MFMA accumulator
+// registers INSIDE this kernel and skip materializing ein/t1/msg/dz1/
+// dz2/dz3 to global (~1.2 KB/edge of stores whose only consumers were
+// the split-K wgrad kernels re-reading them). Costs 68 persistent
+// VGPRs -> 2 waves/SIMD instead of 3; saves the whole wgrad kernel
+// family plus the traffic.
+template <bool FUSE_WG>
+__global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     const bf16* __restrict__ h, const float* __restrict__ coord,
     const float* __restrict__ eattr, const long* __restrict__ row,
     const long* __restrict__ col,
@@ -142,12 +197,19 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
     bf16* __restrict__ dhr_out, bf16* __restrict__ dhc_out,
     float* __restrict__ dcd_out, float* __restrict__ dw3v_out,
     float* __restrict__ gb_out,  // [3*H]: gb1 | gb2 | gb3 column sums
+    float* __restrict__ gw1_out,  // [H][K_OUT] (FUSE_WG only)
+    float* __restrict__ gw2_out,  // [H][H]
+    float* __restrict__ gw3_out,  // [H][H]
     long m, int normalize, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr Smem L = smem_layout();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  // per-wave wgrad accumulators (i-tile = wave), live across the tile loop
+  f32x4 wg1[9] = {};
+  f32x4 wg2[4] = {};
+  f32x4 wg3[4] = {};
 
   float* biases = reinterpret_cast<float*>(smem + L.bias);
   float* wpart = reinterpret_cast<float*>(smem + L.wpart);
@@ -210,13 +272,15 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       for (int k = K_IN; k < K_PAD; ++k) brow[k] = (__bf16)0.f;
     }
     __syncthreads();
-    // ein -> global
-    for (int idx = tid; idx < TILE * (K_OUT / 8); idx += THREADS) {
-      int e = idx / (K_OUT / 8);
-      if (e >= nedge) continue;
-      int c8 = (idx % (K_OUT / 8)) * 8;
-      *reinterpret_cast<bf16x8*>(ein_out + (e0 + e) * K_OUT + c8) =
-          lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
+    // ein -> global (only the split-K wgrad path consumes it)
+    if constexpr (!FUSE_WG) {
+      for (int idx = tid; idx < TILE * (K_OUT / 8); idx += THREADS) {
+        int e = idx / (K_OUT / 8);
+        if (e >= nedge) continue;
+        int c8 = (idx % (K_OUT / 8)) * 8;
+        *reinterpret_cast<bf16x8*>(ein_out + (e0 + e) * K_OUT + c8) =
+            lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
+      }
     }
 
     // ---- recompute: z1, z2, z3 (pre-activations) ----
@@ -237,12 +301,14 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
     }
     __syncthreads();
     // t1 = silu(z1) -> global (coalesced)
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
-          lds8_silu(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+    if constexpr (!FUSE_WG) {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        if (e >= nedge) continue;
+        int c8 = (idx % 8) * 8;
+        *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
+            lds8_silu(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+      }
     }
     {
       f32x4 acc[4] = {};
@@ -260,12 +326,14 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       }
     }
     __syncthreads();
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
-          lds8_silu(smem, L.z2 + (e * H_STRIDE + c8) * 2);
+    if constexpr (!FUSE_WG) {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        if (e >= nedge) continue;
+        int c8 = (idx % 8) * 8;
+        *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
+            lds8_silu(smem, L.z2 + (e * H_STRIDE + c8) * 2);
+      }
     }
     {
       f32x4 acc[4] = {};
@@ -332,12 +400,20 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       atomicAdd(&gbacc[2 * H + c], acc_b3);
     }
     __syncthreads();
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(dz3_out + (e0 + e) * H + c8) =
-          lds8(smem, L.z3 + (e * H_STRIDE + c8) * 2);
+    if constexpr (FUSE_WG) {
+      // dW3 += dz3^T (in z3) @ silu(z2): z3 is consumed below (dmsg
+      // staging overwrites it only after a barrier every wave reaches
+      // after this accumulation)
+      wg_acc<4, true>(smem, L.z3, H_STRIDE * 2, L.z2, H_STRIDE * 2, wave,
+                      lane, wg3);
+    } else {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        if (e >= nedge) continue;
+        int c8 = (idx % 8) * 8;
+        *reinterpret_cast<bf16x8*>(dz3_out + (e0 + e) * H + c8) =
+            lds8(smem, L.z3 + (e * H_STRIDE + c8) * 2);
+      }
     }
 
     // ---- dz2 = (dmsg_n[row] + dz3 @ W3) silu'(z2), overwrite z2 ----
@@ -371,12 +447,19 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       }
     }
     __syncthreads();
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
-          lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
+    if constexpr (FUSE_WG) {
+      // dW2 += dz2^T (in z2) @ silu(z1): z1 still holds pre-activations
+      // (overwritten only after the barrier inside the dz1 phase)
+      wg_acc<4, true>(smem, L.z2, H_STRIDE * 2, L.z1, H_STRIDE * 2, wave,
+                      lane, wg2);
+    } else {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        if (e >= nedge) continue;
+        int c8 = (idx % 8) * 8;
+        *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
+            lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
+      }
     }
     {
       int c = tid & 63;
@@ -407,12 +490,14 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       }
     }
     __syncthreads();
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
-          lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+    if constexpr (!FUSE_WG) {
+      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
+        int e = idx / 8;
+        if (e >= nedge) continue;
+        int c8 = (idx % 8) * 8;
+        *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
+            lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
+      }
     }
     {
       int c = tid & 63;
@@ -422,6 +507,14 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
       for (int e = estart; e < estart + 16; ++e)
         acc_b += (float)z1[e * H_STRIDE + c];
       atomicAdd(&gbacc[c], acc_b);
+    }
+    if constexpr (FUSE_WG) {
+      // dW1 += dz1^T (in z1) @ ein (still in in_tile; K_OUT-col range is
+      // zero-padded past K_IN). Must fully drain before the dein passes
+      // overwrite in_tile -> barrier.
+      wg_acc<9, false>(smem, L.z1, H_STRIDE * 2, L.in_tile, K_STRIDE * 2,
+                       wave, lane, wg1);
+      __syncthreads();
     }
 
     // ---- dein = dz1 @ W1 in 3 register passes of 3 n-tiles ----
@@ -487,11 +580,30 @@ __global__ __launch_bounds__(THREADS, 3) void fused_edge_bwd(
   __syncthreads();
   for (int c = tid; c < H; c += THREADS) atomicAdd(&dw3v_out[c], wpart[c]);
   for (int c = tid; c < 3 * H; c += THREADS) atomicAdd(&gb_out[c], gbacc[c]);
+  if constexpr (FUSE_WG) {
+    // flush wgrad accumulators: D fragment row = (lane>>4)*4 + r,
+    // col = lane & 15; wave owns i-tile `wave`.
+    int i = wave * 16 + (lane >> 4) * 4;
+    int j0 = lane & 15;
+#pragma unroll
+    for (int jt = 0; jt < 9; ++jt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        atomicAdd(&gw1_out[(i + r) * K_OUT + jt * 16 + j0], wg1[jt][r]);
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        atomicAdd(&gw2_out[(i + r) * H + jt * 16 + j0], wg2[jt][r]);
+        atomicAdd(&gw3_out[(i + r) * H + jt * 16 + j0], wg3[jt][r]);
+      }
+  }
 }
 
 }  // namespace
 
-std::vector<torch::Tensor> fused_edge_backward(
+template <bool FUSE_WG>
+static std::vector<torch::Tensor> fused_edge_backward_impl(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
@@ -500,18 +612,25 @@ std::vector<torch::Tensor> fused_edge_backward(
   long m = row.numel();
   auto bopt = h.options();
   auto fopt = coord.options().dtype(torch::kFloat);
-  auto ein = torch::empty({m, (long)K_OUT}, bopt);
-  auto t1 = torch::empty({m, (long)H}, bopt);
-  auto msg = torch::empty({m, (long)H}, bopt);
-  auto dz1 = torch::empty({m, (long)H}, bopt);
-  auto dz2 = torch::empty({m, (long)H}, bopt);
-  auto dz3 = torch::empty({m, (long)H}, bopt);
+  long edge_rows = FUSE_WG ? 0 : m;  // per-edge buffers only when needed
+  auto ein = torch::empty({edge_rows, (long)K_OUT}, bopt);
+  auto t1 = torch::empty({edge_rows, (long)H}, bopt);
+  auto msg = torch::empty({edge_rows, (long)H}, bopt);
+  auto dz1 = torch::empty({edge_rows, (long)H}, bopt);
+  auto dz2 = torch::empty({edge_rows, (long)H}, bopt);
+  auto dz3 = torch::empty({edge_rows, (long)H}, bopt);
   auto dhr = torch::empty({m, (long)H}, bopt);
   auto dhc = torch::empty({m, (long)H}, bopt);
   auto dcd = torch::empty({m, 3}, fopt);
   auto dw3v = torch::zeros({(long)H}, fopt);
   auto gb = torch::zeros({3 * (long)H}, fopt);
-  if (m == 0) return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
+  auto gw1 = torch::zeros({FUSE_WG ? (long)H : 0, (long)K_OUT}, fopt);
+  auto gw2 = torch::zeros({FUSE_WG ? (long)H : 0, (long)H}, fopt);
+  auto gw3 = torch::zeros({FUSE_WG ? (long)H : 0, (long)H}, fopt);
+  if (m == 0) {
+    if (FUSE_WG) return {dhr, dhc, dcd, dw3v, gb, gw1, gw2, gw3};
+    return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
+  }
   auto stream = at::hip::getCurrentHIPStream();
   constexpr Smem L = smem_layout();
   long tiles = (m + TILE - 1) / TILE;
@@ -533,7 +652,7 @@ std::vector<torch::Tensor> fused_edge_backward(
   auto b2c = b2.contiguous().to(torch::kFloat);
   auto b3c = b3.contiguous().to(torch::kFloat);
   auto w3vc = w3v.contiguous().to(torch::kFloat);
-  fused_edge_bwd<<<blocks, THREADS, L.total, stream>>>(
+  fused_edge_bwd<FUSE_WG><<<blocks, THREADS, L.total, stream>>>(
       reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
       col.contiguous().data_ptr<long>(),
@@ -553,7 +672,36 @@ std::vector<torch::Tensor> fused_edge_backward(
       reinterpret_cast<bf16*>(dz3.data_ptr()),
       reinterpret_cast<bf16*>(dhr.data_ptr()),
       reinterpret_cast<bf16*>(dhc.data_ptr()), dcd.data_ptr<float>(),
-      dw3v.data_ptr<float>(), gb.data_ptr<float>(), m, normalize ? 1 : 0,
+      dw3v.data_ptr<float>(), gb.data_ptr<float>(),
+      FUSE_WG ? gw1.data_ptr<float>() : nullptr,
+      FUSE_WG ? gw2.data_ptr<float>() : nullptr,
+      FUSE_WG ? gw3.data_ptr<float>() : nullptr, m, normalize ? 1 : 0,
       (float)eps);
+  if (FUSE_WG) return {dhr, dhc, dcd, dw3v, gb, gw1, gw2, gw3};
   return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
+}
+
+std::vector<torch::Tensor> fused_edge_backward(
+    torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
+    torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
+    torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    torch::Tensor w3v, bool normalize, double eps) {
+  return fused_edge_backward_impl<false>(h, coord, eattr, row, col, dmsg_n,
+                                         dtrans_n, w1, b1, w2, b2, w3, b3,
+                                         w3v, normalize, eps);
+}
+
+// wgrad-fused variant: returns {dhr, dhc, dcd, dw3v, gb, gw1, gw2, gw3} —
+// the three weight gradients come out of the kernel directly; no per-edge
+// intermediates are materialized.
+std::vector<torch::Tensor> fused_edge_backward_wg(
+    torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
+    torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
+    torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    torch::Tensor w3v, bool normalize, double eps) {
+  return fused_edge_backward_impl<true>(h, coord, eattr, row, col, dmsg_n,
+                                        dtrans_n, w1, b1, w2, b2, w3, b3,
+                                        w3v, normalize, eps);
 }
