@@ -135,6 +135,11 @@ def train_step(graphed, batch, optimizer, grad_bucket, step, accum,
                                                max_norm=0.3)
             optimizer.step()
             optimizer.zero_grad(set_to_none=False)
+            # captured replays read version-cached weight transforms;
+            # refresh them after the weights changed (ops/prep.py)
+            from distegnn_amd import ops
+
+            ops.refresh_weight_prep()
 
         # side stream: eager allocs must not alias graph-pool blocks
         graphed.run_eager(_opt)

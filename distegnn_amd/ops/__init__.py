@@ -16,6 +16,7 @@ from typing import Optional
 import torch
 
 from . import reference
+from .prep import refresh as refresh_weight_prep  # noqa: F401
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
@@ -294,9 +295,15 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
     def forward(ctx, h, coord, eattr, row, col, rowptr, colptr, col_perm,
                 w1, b1, w2, b2, w3, b3, w3v, normalize, eps):
         ext = _require_ext("fused_edge_block")
+        from . import prep
+
+        prepped = [prep.get(w1, "pad_kpad"), prep.get(w2, "bf16"),
+                   prep.get(w3, "bf16"), prep.get(b1, "f32"),
+                   prep.get(b2, "f32"), prep.get(b3, "f32"),
+                   prep.get(w3v, "f32")]
         msg, trans = ext.fused_edge_forward(
-            h, coord, eattr, row, col, w1.bfloat16(), b1, w2.bfloat16(), b2,
-            w3.bfloat16(), b3, w3v, bool(normalize), float(eps))
+            h, coord, eattr, row, col, w1, b1, w2, b2,
+            w3, b3, w3v, bool(normalize), float(eps), prepped)
         agg_msg = ext.segment_reduce_csr(msg, rowptr, True)
         agg_trans = ext.segment_reduce_csr(trans, rowptr, True)
         ctx.save_for_backward(h, coord, eattr, row, col, rowptr, colptr,
@@ -318,9 +325,16 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             dtrans_n = (dagg_trans
                         / deg.unsqueeze(-1).to(dagg_trans.dtype)).contiguous()
             k_in = w1.size(1)
+            from . import prep
+
+            prepped = [prep.get(w1, "pad_kpad"), prep.get(w1, "tpad_kout"),
+                       prep.get(w2, "bf16"), prep.get(w2, "t_bf16"),
+                       prep.get(w3, "bf16"), prep.get(w3, "t_bf16"),
+                       prep.get(b1, "f32"), prep.get(b2, "f32"),
+                       prep.get(b3, "f32"), prep.get(w3v, "f32")]
             args = (h, coord, eattr, row, col, dmsg_n, dtrans_n,
-                    w1.bfloat16(), b1, w2.bfloat16(), b2, w3.bfloat16(), b3,
-                    w3v, bool(ctx.normalize), float(ctx.eps))
+                    w1, b1, w2, b2, w3, b3,
+                    w3v, bool(ctx.normalize), float(ctx.eps), prepped)
             # measured on MI355X (gpurun_out/r2_call2.log): the wg-fused
             # kernel eliminates ~2.5 ms/step of split-K wgrad work but its
             # 68 persistent accumulator VGPRs drop occupancy 3 -> 2
@@ -485,10 +499,17 @@ class _FusedVirtualBlockFn(torch.autograd.Function):
                 chunks_ce, chunks_scp, w1, b1, w2, b2, wxv, bxv, wxvv, wX,
                 bX, wXv, train):
         ext = _require_ext("fused_virtual_block")
+        from . import prep
+
+        prepped = [prep.get(w1, "pad_kpad"), prep.get(w2, "bf16"),
+                   prep.get(wxv, "bf16"), prep.get(wX, "bf16"),
+                   prep.get(b1, "f32"), prep.get(b2, "f32"),
+                   prep.get(bxv, "f32"), prep.get(bX, "f32"),
+                   prep.get(wxvv, "f32"), prep.get(wXv, "f32")]
         outs = ext.fused_virtual_forward(
             h, coord, vcoord.float(), vfeat.bfloat16(), gram.float(), batch,
-            w1.bfloat16(), b1, w2.bfloat16(), b2, wxv.bfloat16(), bxv, wxvv,
-            wX.bfloat16(), bX, wXv, train)
+            w1, b1, w2, b2, wxv, bxv, wxvv,
+            wX, bX, wXv, train, prepped)
         vmsg, tv, tx, vin, z1, z2, zxv, zX, p2 = outs
         n, c = h.size(0), vcoord.size(1)
         ctx.save_for_backward(h, coord, vcoord, batch, ptr, chunks_cb,
@@ -505,14 +526,19 @@ class _FusedVirtualBlockFn(torch.autograd.Function):
         ext = _load_extension()
         n, c = ctx.shape_nc
         rows = n * c
+        from . import prep
+
+        prepped = [prep.get(w1, "tpad_kout"), prep.get(w2, "t_bf16"),
+                   prep.get(wxv, "t_bf16"), prep.get(wX, "t_bf16"),
+                   prep.get(wxvv, "f32"), prep.get(wXv, "f32")]
         (dz1, dz2, dzxv, dzX, dh_row, dvf_row, dgram_row, dvd,
          dp2, gb) = ext.fused_virtual_backward(
             coord, vcoord.float(), batch,
             dvmsg.reshape(rows, -1).to(torch.bfloat16).contiguous(),
             dtv.reshape(rows, 3).float().contiguous(),
             dtx.reshape(rows, 3).float().contiguous(),
-            z1, z2, zxv, zX, p2, w1.bfloat16(), w2.bfloat16(),
-            wxv.bfloat16(), wX.bfloat16(), wxvv, wXv)
+            z1, z2, zxv, zX, p2, w1, w2,
+            wxv, wX, wxvv, wXv, prepped)
         from .linear import chunked_wgrad
 
         k_in = w1.size(1)
@@ -596,5 +622,5 @@ __all__ = [
     "segment_sum", "segment_mean", "graph_sum_pool", "graph_mean_pool",
     "gather_rows", "fused_edge_block", "eager_edge_block",
     "fused_virtual_block", "eager_virtual_block", "radius_graph",
-    "hip_ext", "reference",
+    "hip_ext", "reference", "refresh_weight_prep",
 ]

@@ -19,7 +19,8 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps);
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped = {});
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
 torch::Tensor tall_linear(torch::Tensor x, torch::Tensor bmat,
@@ -29,25 +30,29 @@ std::vector<torch::Tensor> fused_virtual_forward(
     torch::Tensor vfeat, torch::Tensor gram, torch::Tensor batch,
     torch::Tensor w1, torch::Tensor b1, torch::Tensor w2, torch::Tensor b2,
     torch::Tensor wxv, torch::Tensor bxv, torch::Tensor wxvv,
-    torch::Tensor wX, torch::Tensor bX, torch::Tensor wXv, bool train);
+    torch::Tensor wX, torch::Tensor bX, torch::Tensor wXv, bool train,
+    std::vector<torch::Tensor> prepped = {});
 std::vector<torch::Tensor> fused_virtual_backward(
     torch::Tensor coord, torch::Tensor vcoord, torch::Tensor batch,
     torch::Tensor dvmsg, torch::Tensor dtv, torch::Tensor dtx,
     torch::Tensor z1, torch::Tensor z2, torch::Tensor zxv, torch::Tensor zX,
     torch::Tensor p2, torch::Tensor w1, torch::Tensor w2, torch::Tensor wxv,
-    torch::Tensor wX, torch::Tensor wxvv, torch::Tensor wXv);
+    torch::Tensor wX, torch::Tensor wxvv, torch::Tensor wXv,
+    std::vector<torch::Tensor> prepped = {});
 std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps);
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped = {});
 std::vector<torch::Tensor> fused_edge_backward_wg(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps);
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped = {});
 
 torch::Tensor radius_graph(torch::Tensor pos, double r) {
   return std::get<0>(radius_graph_gpu(pos, r));
@@ -84,14 +89,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("h"), py::arg("coord"), py::arg("eattr"), py::arg("row"),
         py::arg("col"), py::arg("w1"), py::arg("b1"), py::arg("w2"),
         py::arg("b2"), py::arg("w3"), py::arg("b3"), py::arg("w3v"),
-        py::arg("normalize"), py::arg("eps"));
+        py::arg("normalize"), py::arg("eps"),
+        py::arg("prepped") = std::vector<torch::Tensor>{});
   m.def("fused_edge_backward", &fused_edge_backward,
         "fused edge-block backward: in-LDS recompute + per-edge grads",
         py::arg("h"), py::arg("coord"), py::arg("eattr"), py::arg("row"),
         py::arg("col"), py::arg("dmsg_n"), py::arg("dtrans_n"),
         py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
         py::arg("w3"), py::arg("b3"), py::arg("w3v"), py::arg("normalize"),
-        py::arg("eps"));
+        py::arg("eps"),
+        py::arg("prepped") = std::vector<torch::Tensor>{});
   m.def("fused_edge_backward_wg", &fused_edge_backward_wg,
         "edge-block backward with in-kernel MFMA weight gradients "
         "(no per-edge intermediates): {dhr, dhc, dcd, dw3v, gb, gw1, "
@@ -100,7 +107,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("col"), py::arg("dmsg_n"), py::arg("dtrans_n"),
         py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
         py::arg("w3"), py::arg("b3"), py::arg("w3v"), py::arg("normalize"),
-        py::arg("eps"));
+        py::arg("eps"),
+        py::arg("prepped") = std::vector<torch::Tensor>{});
   m.def("wgrad_splitk", &wgrad_splitk_launch,
         "split-K MFMA weight gradient: g^T @ x for tall activations",
         py::arg("g"), py::arg("x"));

@@ -276,7 +276,8 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps) {
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped) {
   TORCH_CHECK(h.is_cuda() && h.scalar_type() == torch::kBFloat16,
               "h must be CUDA bf16");
   TORCH_CHECK(h.size(1) == H, "fused edge kernel requires hidden_nf=64");
@@ -292,14 +293,25 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
   constexpr SmemLayout L = smem_layout();
   long tiles = (m + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
-  // row-pad W1 to [64][K_PAD] so 16-B B-fragment reads are aligned
-  auto w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - K_IN});
-  auto w2c = w2.contiguous();
-  auto w3c = w3.contiguous();
-  auto b1c = b1.contiguous().to(torch::kFloat);
-  auto b2c = b2.contiguous().to(torch::kFloat);
-  auto b3c = b3.contiguous().to(torch::kFloat);
-  auto w3vc = w3v.contiguous().to(torch::kFloat);
+  // Weight prep: python may pass version-cached transformed copies
+  // (ops/prep.py) so replays/graphs skip ~7 tiny cast/pad kernels per
+  // call; otherwise transform here.
+  torch::Tensor w1p, w2c, w3c, b1c, b2c, b3c, w3vc;
+  if (!prepped.empty()) {
+    TORCH_CHECK(prepped.size() == 7, "edge fwd prepped wants 7 tensors");
+    w1p = prepped[0]; w2c = prepped[1]; w3c = prepped[2];
+    b1c = prepped[3]; b2c = prepped[4]; b3c = prepped[5]; w3vc = prepped[6];
+    TORCH_CHECK(w1p.size(1) == K_PAD, "prepped w1p must be row-padded");
+  } else {
+    // row-pad W1 to [64][K_PAD] so 16-B B-fragment reads are aligned
+    w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - K_IN});
+    w2c = w2.contiguous();
+    w3c = w3.contiguous();
+    b1c = b1.contiguous().to(torch::kFloat);
+    b2c = b2.contiguous().to(torch::kFloat);
+    b3c = b3.contiguous().to(torch::kFloat);
+    w3vc = w3v.contiguous().to(torch::kFloat);
+  }
   fused_edge_fwd<<<blocks, THREADS, L.total, stream>>>(
       reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),

@@ -608,7 +608,8 @@ static std::vector<torch::Tensor> fused_edge_backward_impl(
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps) {
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped) {
   long m = row.numel();
   auto bopt = h.options();
   auto fopt = coord.options().dtype(torch::kFloat);
@@ -640,18 +641,28 @@ static std::vector<torch::Tensor> fused_edge_backward_impl(
   auto ec = eattr.contiguous().to(torch::kFloat);
   auto dmn = dmsg_n.contiguous();
   auto dtn = dtrans_n.contiguous().to(torch::kFloat);
-  auto w1c = w1.contiguous();
-  auto w1p = torch::constant_pad_nd(w1c, {0, K_PAD - K_IN});
-  auto w1tp = torch::constant_pad_nd(w1c.t().contiguous(),
-                                     {0, 0, 0, K_OUT - K_IN});
-  auto w2c = w2.contiguous();
-  auto w2tc = w2c.t().contiguous();
-  auto w3c = w3.contiguous();
-  auto w3tc = w3c.t().contiguous();
-  auto b1c = b1.contiguous().to(torch::kFloat);
-  auto b2c = b2.contiguous().to(torch::kFloat);
-  auto b3c = b3.contiguous().to(torch::kFloat);
-  auto w3vc = w3v.contiguous().to(torch::kFloat);
+  torch::Tensor w1p, w1tp, w2c, w2tc, w3c, w3tc, b1c, b2c, b3c, w3vc;
+  if (!prepped.empty()) {
+    TORCH_CHECK(prepped.size() == 10, "edge bwd prepped wants 10 tensors");
+    w1p = prepped[0]; w1tp = prepped[1]; w2c = prepped[2];
+    w2tc = prepped[3]; w3c = prepped[4]; w3tc = prepped[5];
+    b1c = prepped[6]; b2c = prepped[7]; b3c = prepped[8]; w3vc = prepped[9];
+    TORCH_CHECK(w1p.size(1) == K_PAD && w1tp.size(0) == K_OUT,
+                "prepped w1p/w1tp must be padded");
+  } else {
+    auto w1c = w1.contiguous();
+    w1p = torch::constant_pad_nd(w1c, {0, K_PAD - K_IN});
+    w1tp = torch::constant_pad_nd(w1c.t().contiguous(),
+                                  {0, 0, 0, K_OUT - K_IN});
+    w2c = w2.contiguous();
+    w2tc = w2c.t().contiguous();
+    w3c = w3.contiguous();
+    w3tc = w3c.t().contiguous();
+    b1c = b1.contiguous().to(torch::kFloat);
+    b2c = b2.contiguous().to(torch::kFloat);
+    b3c = b3.contiguous().to(torch::kFloat);
+    w3vc = w3v.contiguous().to(torch::kFloat);
+  }
   fused_edge_bwd<FUSE_WG><<<blocks, THREADS, L.total, stream>>>(
       reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
@@ -686,10 +697,12 @@ std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps) {
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped) {
   return fused_edge_backward_impl<false>(h, coord, eattr, row, col, dmsg_n,
                                          dtrans_n, w1, b1, w2, b2, w3, b3,
-                                         w3v, normalize, eps);
+                                         w3v, normalize, eps,
+                                         std::move(prepped));
 }
 
 // wgrad-fused variant: returns {dhr, dhc, dcd, dw3v, gb, gw1, gw2, gw3} —
@@ -700,8 +713,10 @@ std::vector<torch::Tensor> fused_edge_backward_wg(
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
     torch::Tensor dtrans_n, torch::Tensor w1, torch::Tensor b1,
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
-    torch::Tensor w3v, bool normalize, double eps) {
+    torch::Tensor w3v, bool normalize, double eps,
+    std::vector<torch::Tensor> prepped) {
   return fused_edge_backward_impl<true>(h, coord, eattr, row, col, dmsg_n,
                                         dtrans_n, w1, b1, w2, b2, w3, b3,
-                                        w3v, normalize, eps);
+                                        w3v, normalize, eps,
+                                        std::move(prepped));
 }

@@ -610,7 +610,8 @@ std::vector<torch::Tensor> fused_virtual_forward(
     torch::Tensor vfeat, torch::Tensor gram, torch::Tensor batch,
     torch::Tensor w1, torch::Tensor b1, torch::Tensor w2, torch::Tensor b2,
     torch::Tensor wxv, torch::Tensor bxv, torch::Tensor wxvv,
-    torch::Tensor wX, torch::Tensor bX, torch::Tensor wXv, bool train) {
+    torch::Tensor wX, torch::Tensor bX, torch::Tensor wXv, bool train,
+    std::vector<torch::Tensor> prepped) {
   TORCH_CHECK(h.is_cuda() && h.scalar_type() == torch::kBFloat16,
               "h must be CUDA bf16");
   TORCH_CHECK(h.size(1) == H, "fused virtual kernel requires hidden_nf=64");
@@ -636,19 +637,28 @@ std::vector<torch::Tensor> fused_virtual_forward(
   constexpr VSmem L = vsmem_layout();
   long tiles = (rows + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
-  auto w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - k_in});
+  torch::Tensor w1p, w2c, wxvc, wXc, b1c, b2c, bxvc, bXc, wxvvc, wXvc;
+  if (!prepped.empty()) {
+    TORCH_CHECK(prepped.size() == 10, "virtual fwd prepped wants 10");
+    w1p = prepped[0]; w2c = prepped[1]; wxvc = prepped[2]; wXc = prepped[3];
+    b1c = prepped[4]; b2c = prepped[5]; bxvc = prepped[6]; bXc = prepped[7];
+    wxvvc = prepped[8]; wXvc = prepped[9];
+    TORCH_CHECK(w1p.size(1) == K_PAD, "prepped w1p must be row-padded");
+  } else {
+    w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - k_in});
+    w2c = w2.contiguous(); wxvc = wxv.contiguous(); wXc = wX.contiguous();
+    b1c = b1.contiguous().to(torch::kFloat);
+    b2c = b2.contiguous().to(torch::kFloat);
+    bxvc = bxv.contiguous().to(torch::kFloat);
+    bXc = bX.contiguous().to(torch::kFloat);
+    wxvvc = wxvv.contiguous().to(torch::kFloat);
+    wXvc = wXv.contiguous().to(torch::kFloat);
+  }
   auto hc = h.contiguous();
   auto cc_ = coord.contiguous().to(torch::kFloat);
   auto vc = vcoord.contiguous().to(torch::kFloat);
   auto vf = vfeat.contiguous();
   auto gr = gram.contiguous().to(torch::kFloat);
-  auto w2c = w2.contiguous(), wxvc = wxv.contiguous(), wXc = wX.contiguous();
-  auto b1c = b1.contiguous().to(torch::kFloat);
-  auto b2c = b2.contiguous().to(torch::kFloat);
-  auto bxvc = bxv.contiguous().to(torch::kFloat);
-  auto bXc = bX.contiguous().to(torch::kFloat);
-  auto wxvvc = wxvv.contiguous().to(torch::kFloat);
-  auto wXvc = wXv.contiguous().to(torch::kFloat);
   auto bp = batch.contiguous();
 
 #define ARGS                                                                 \
@@ -684,7 +694,8 @@ std::vector<torch::Tensor> fused_virtual_backward(
     torch::Tensor dvmsg, torch::Tensor dtv, torch::Tensor dtx,
     torch::Tensor z1, torch::Tensor z2, torch::Tensor zxv, torch::Tensor zX,
     torch::Tensor p2, torch::Tensor w1, torch::Tensor w2, torch::Tensor wxv,
-    torch::Tensor wX, torch::Tensor wxvv, torch::Tensor wXv) {
+    torch::Tensor wX, torch::Tensor wxvv, torch::Tensor wXv,
+    std::vector<torch::Tensor> prepped) {
   long rows = z1.size(0);
   int cdim = (int)vcoord.size(1);
   int k_in = 2 * H + 1 + cdim;
@@ -706,14 +717,22 @@ std::vector<torch::Tensor> fused_virtual_backward(
   constexpr VSmem L = vsmem_layout();
   long tiles = (rows + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
-  auto w1c = w1.contiguous();
-  auto w1tp = torch::constant_pad_nd(w1c.t().contiguous(),
-                                     {0, 0, 0, K_OUT - k_in});
-  auto w2tc = w2.contiguous().t().contiguous();
-  auto wxvtc = wxv.contiguous().t().contiguous();
-  auto wXtc = wX.contiguous().t().contiguous();
-  auto wxvvc = wxvv.contiguous().to(torch::kFloat);
-  auto wXvc = wXv.contiguous().to(torch::kFloat);
+  torch::Tensor w1tp, w2tc, wxvtc, wXtc, wxvvc, wXvc;
+  if (!prepped.empty()) {
+    TORCH_CHECK(prepped.size() == 6, "virtual bwd prepped wants 6");
+    w1tp = prepped[0]; w2tc = prepped[1]; wxvtc = prepped[2];
+    wXtc = prepped[3]; wxvvc = prepped[4]; wXvc = prepped[5];
+    TORCH_CHECK(w1tp.size(0) == K_OUT, "prepped w1tp must be row-padded");
+  } else {
+    auto w1c = w1.contiguous();
+    w1tp = torch::constant_pad_nd(w1c.t().contiguous(),
+                                  {0, 0, 0, K_OUT - k_in});
+    w2tc = w2.contiguous().t().contiguous();
+    wxvtc = wxv.contiguous().t().contiguous();
+    wXtc = wX.contiguous().t().contiguous();
+    wxvvc = wxvv.contiguous().to(torch::kFloat);
+    wXvc = wXv.contiguous().to(torch::kFloat);
+  }
   auto cc_ = coord.contiguous().to(torch::kFloat);
   auto vc = vcoord.contiguous().to(torch::kFloat);
   fused_virtual_bwd<<<blocks, THREADS, L.zd, stream>>>(

@@ -60,31 +60,55 @@ def _ext_ok(x, w):
             and w.size(0) <= 208 and w.size(1) <= 208)
 
 
+def _cast(t, dt):
+    """Compute-dtype cast, version-cached for bf16 weights (ops/prep.py):
+    under captured hipGraphs the per-call .to(bf16) copies replay every
+    step; the cache turns them into reads of refreshed buffers."""
+    if t is None or dt is None or t.dtype == dt:
+        return t
+    if dt == torch.bfloat16 and t.is_cuda:
+        from . import prep
+
+        return prep.get(t, "bf16")
+    return t.to(dt)
+
+
 class _SplitKLinearFn(torch.autograd.Function):
+    """Takes the RAW parameters (grads returned in parameter dtype); the
+    compute-dtype cast happens inside so it can be version-cached."""
+
     @staticmethod
-    def forward(ctx, x, w, b):
+    def forward(ctx, x, w, b, dt):
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
-        if _ext_ok(x, w):
+        ctx.dt = dt
+        wc = _cast(w, dt)
+        bc = _cast(b, dt)
+        if _ext_ok(x, wc):
             from . import hip_ext
 
-            return hip_ext().tall_linear(x, w, b, 0)
-        return F.linear(x, w, b)
+            return hip_ext().tall_linear(x, wc, bc, 0)
+        return F.linear(x, wc, bc)
 
     @staticmethod
     def backward(ctx, g):
         x, w = ctx.saved_tensors
         g = g.contiguous()
-        wt = w.to(g.dtype)
-        if _ext_ok(g, wt.t()):
+        if ctx.dt == torch.bfloat16 and w.is_cuda and w.dtype != ctx.dt:
+            from . import prep
+
+            wt = prep.get(w, "t_bf16")
+        else:
+            wt = w.to(g.dtype).t().contiguous()
+        if _ext_ok(g, wt):
             from . import hip_ext
 
-            gx = hip_ext().tall_linear(g, wt.t().contiguous(), None, 0)
+            gx = hip_ext().tall_linear(g, wt, None, 0)
         else:
-            gx = g @ wt
+            gx = g @ wt.t()
         gw = chunked_wgrad(g, x.to(g.dtype)).to(w.dtype)
         gb = g.sum(0).to(w.dtype) if ctx.has_bias else None
-        return gx, gw, gb
+        return gx, gw, gb, None
 
 
 class SplitKLinear(nn.Linear):
@@ -95,17 +119,15 @@ class SplitKLinear(nn.Linear):
                 and x.numel() // x.size(-1) >= _MIN_ROWS):
             shape = x.shape
             x2 = x.reshape(-1, shape[-1])
-            w = self.weight
-            b = self.bias
             if torch.is_autocast_enabled():
                 dt = torch.get_autocast_dtype("cuda")
                 x2 = x2.to(dt)
-                w = w.to(dt)
-                b = b.to(dt) if b is not None else None
-            elif x2.dtype != w.dtype and x2.dtype in (torch.bfloat16,
-                                                      torch.float16):
-                w = w.to(x2.dtype)
-                b = b.to(x2.dtype) if b is not None else None
-            out = _SplitKLinearFn.apply(x2.contiguous(), w, b)
+            elif x2.dtype != self.weight.dtype and x2.dtype in (
+                    torch.bfloat16, torch.float16):
+                dt = x2.dtype
+            else:
+                dt = None
+            out = _SplitKLinearFn.apply(x2.contiguous(), self.weight,
+                                        self.bias, dt)
             return out.reshape(*shape[:-1], out.size(-1))
         return super().forward(x)

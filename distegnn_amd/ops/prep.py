@@ -1,0 +1,108 @@
+"""Version-cached weight transforms for the fused HIP kernels.
+
+The fused edge/virtual kernels read weights from padded/transposed/casted
+global copies (bf16 K-contiguous fragments, fp32 biases — docs/KERNELS.md).
+Round 1 recomputed those per call: ~150 tiny cast/pad/transpose kernels per
+training step, replayed inside the captured hipGraph (~0.7-1.2 ms/step of
+pure launch-sized kernels; profiles/README.md roadmap #3).
+
+This cache keys each transform on the source tensor's ``_version`` counter:
+
+* eager path: ``get`` checks the version per call and recomputes in place
+  (``copy_`` into the existing buffer — stable addresses) only after the
+  optimizer actually stepped.
+* captured path: at capture time the cache is warm, so the graph records
+  only READS of the cached buffers. Replays never run host code, therefore
+  the runtime must call :func:`refresh` after every optimizer step (the
+  trainer/bench optimizer regions do; see runtime/trainer.py) — it
+  recomputes stale entries into the SAME buffers the graph references.
+
+No-op transforms (e.g. ``.float()`` of an fp32 bias) are detected by
+storage identity and passed through uncached.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Dict, Tuple
+
+import torch
+import torch.nn.functional as F
+
+K_PAD = 160   # row padding of W1 for aligned 16-B B-fragments
+K_OUT = 144   # row count of the padded W1^T used by dgrad
+
+
+def bf16c(t: torch.Tensor) -> torch.Tensor:
+    return t.detach().bfloat16().contiguous()
+
+
+def f32c(t: torch.Tensor) -> torch.Tensor:
+    return t.detach().float().contiguous()
+
+
+def pad_kpad(t: torch.Tensor) -> torch.Tensor:
+    """[O, k] -> [O, K_PAD] bf16 (column zero-pad)."""
+    return F.pad(t.detach().bfloat16(), (0, K_PAD - t.size(1))).contiguous()
+
+
+def tpad_kout(t: torch.Tensor) -> torch.Tensor:
+    """[O, k] -> [K_OUT, O] bf16 (transpose, row zero-pad)."""
+    tt = t.detach().bfloat16().t().contiguous()
+    return F.pad(tt, (0, 0, 0, K_OUT - tt.size(0))).contiguous()
+
+
+def t_bf16(t: torch.Tensor) -> torch.Tensor:
+    return t.detach().bfloat16().t().contiguous()
+
+
+_TRANSFORMS: Dict[str, Callable[[torch.Tensor], torch.Tensor]] = {
+    "bf16": bf16c, "f32": f32c, "pad_kpad": pad_kpad,
+    "tpad_kout": tpad_kout, "t_bf16": t_bf16,
+}
+
+# (id(tensor), kind) -> [version, buffer, source-tensor-ref]
+_CACHE: Dict[Tuple[int, str], list] = {}
+
+
+def get(t: torch.Tensor, kind: str) -> torch.Tensor:
+    fn = _TRANSFORMS[kind]
+    if not t.is_cuda:
+        return fn(t)                      # CPU path: no caching needed
+    key = (id(t), kind)
+    ent = _CACHE.get(key)
+    v = t._version
+    if ent is None:
+        buf = fn(t)
+        if buf.data_ptr() == t.data_ptr() and buf.dtype == t.dtype:
+            return buf                    # no-op transform: don't cache
+        _CACHE[key] = [v, buf, t]         # keep t alive (id stability)
+        return buf
+    if ent[0] != v:
+        ent[1].copy_(fn(t))
+        ent[0] = v
+    return ent[1]
+
+
+def refresh() -> int:
+    """Recompute every stale cached transform in place. MUST be called
+    after optimizer steps when training under captured hipGraphs (replays
+    read the cached buffers without running this module). Returns the
+    number of refreshed entries."""
+    n = 0
+    for (tid, kind), ent in _CACHE.items():
+        t = ent[2]
+        if ent[0] != t._version:
+            ent[1].copy_(_TRANSFORMS[kind](t))
+            ent[0] = t._version
+            n += 1
+    return n
+
+
+def any_stale() -> bool:
+    """Host-only version scan (no kernels) — lets GraphedStep decide
+    whether a side-stream refresh is needed before replaying."""
+    return any(ent[0] != ent[2]._version for ent in _CACHE.values())
+
+
+def clear():
+    _CACHE.clear()

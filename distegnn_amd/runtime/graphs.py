@@ -150,6 +150,15 @@ class GraphedStep:
     def __call__(self, batch):
         if not self.enabled:
             return self.step_fn(batch)
+        # captured graphs read version-cached weight transforms
+        # (ops/prep.py); if the optimizer stepped since the last refresh,
+        # refresh them on the side stream BEFORE replaying — replaying
+        # against stale weights would silently train wrong. Host-only
+        # version scan when nothing changed.
+        from ..ops import prep as _prep
+
+        if _prep.any_stale():
+            self.run_eager(_prep.refresh)
         key = self._key(batch)
         e = self.entries.get(key)
         if e is None:

@@ -47,6 +47,12 @@ def _restore_state(params, optimizer, snap):
                 p.grad.zero_()
     if optimizer is not None and opt_sd is not None:
         optimizer.load_state_dict(copy.deepcopy(opt_sd))
+    # restoring parameters bumps their version counters: cached weight
+    # transforms (ops/prep.py) referenced by captured graphs must be
+    # refreshed in the same region
+    from ..ops import refresh_weight_prep
+
+    refresh_weight_prep()
 
 
 def run_capture_integrity_gate(

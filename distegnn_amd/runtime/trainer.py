@@ -263,6 +263,9 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
                     if scheduler is not None:
                         scheduler.step()
                     optimizer.zero_grad(set_to_none=False)
+                    from .. import ops
+
+                    ops.refresh_weight_prep()
 
                 if graphed_step is not None:
                     # side stream: eager optimizer allocs must not alias
@@ -424,6 +427,9 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                                                      max_norm=0.3)
                         optimizer.step()
                         optimizer.zero_grad(set_to_none=False)
+                        from .. import ops
+
+                        ops.refresh_weight_prep()
 
                     graphed_step.run_eager(_opt)
                 return mse_log
