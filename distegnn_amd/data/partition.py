@@ -172,12 +172,18 @@ def graph_partition(edge_index: torch.Tensor, num_nodes: int, num_parts: int,
                     seed: int = 0) -> torch.Tensor:
     """Balanced k-way node partition of a graph (METIS-mode back-end).
 
-    Recursive bisection: each bisection grows one half by BFS from a
-    geometric extreme (the node furthest along the current cloud's principal
-    axis when ``pos`` is given, else node 0), stopping at half the target
-    size. BFS growth keeps parts connected/edge-local (the METIS objective
-    that matters here: minimize cut real edges, which DistEGNN drops), and
-    exact size targets keep partitions balanced. Deterministic.
+    With ``pos`` (the DistEGNN case — radius graphs ARE geometric):
+    recursive coordinate bisection, median split along each subcloud's
+    principal axis. Cuts are near-planar, so for a radius graph the cut
+    fraction approaches the geometric optimum (measured on the 113K
+    synthetic cloud: 2.0%/5.3%/8.7% at ws=2/4/8 vs 3.8%/7.8%/11.2% for
+    kmeans and ~50-87% for random — profiles/partition_quality.json),
+    partitions are EXACTLY balanced (better than METIS's 1.03 tolerance)
+    and the result is deterministic, so the preprocessing cache key stays
+    meaningful (SURVEY.md §7 hard-part 7).
+
+    Without ``pos``: bisection grows one half by BFS from node 0 —
+    topology-only fallback with the same balance guarantee.
     """
     order = np.arange(num_nodes)
     labels = np.zeros(num_nodes, dtype=np.int64)
@@ -197,13 +203,15 @@ def graph_partition(edge_index: torch.Tensor, num_nodes: int, num_parts: int,
             return
         left_parts = parts // 2
         target_left = int(round(len(nodes) * left_parts / parts))
-        inset = np.zeros(num_nodes, dtype=bool)
-        inset[nodes] = True
-        # seed: geometric extreme along the principal axis of this cloud
         if p is not None and len(nodes) > 1:
+            # Recursive coordinate bisection: median split along the best
+            # of {x, y, z, principal axis} — chosen by the ACTUAL edge cut
+            # each candidate plane produces on this subgraph (an isotropic
+            # cloud's principal axis degenerates to a diagonal, whose cut
+            # plane is up to sqrt(3)x larger than an axis-aligned one).
+            # Near-planar cuts, exact balance, deterministic.
             sub = p[nodes]
             centered = sub - sub.mean(0)
-            # principal axis via power iteration on the 3x3 covariance
             cov = centered.T @ centered
             v = np.ones(cov.shape[0])
             for _ in range(16):
@@ -212,9 +220,30 @@ def graph_partition(edge_index: torch.Tensor, num_nodes: int, num_parts: int,
                 if nv == 0:
                     break
                 v /= nv
-            seed_node = nodes[int(np.argmin(centered @ v))]
-        else:
-            seed_node = nodes[0]
+            inset = np.zeros(num_nodes, dtype=bool)
+            inset[nodes] = True
+            sub_e = inset[row_s] & inset[col_s]
+            er, ec = row_s[sub_e], col_s[sub_e]
+            best = None
+            cands = [centered[:, d] for d in range(centered.shape[1])]
+            cands.append(centered @ v)
+            side = np.zeros(num_nodes, dtype=bool)
+            for proj in cands:
+                order_ax = np.argsort(proj, kind="stable")
+                side[nodes] = False
+                side[nodes[order_ax[:target_left]]] = True
+                cut = int(np.count_nonzero(side[er] != side[ec]))
+                if best is None or cut < best[0]:
+                    best = (cut, order_ax)
+            order_ax = best[1]
+            left = nodes[order_ax[:target_left]]
+            right = nodes[order_ax[target_left:]]
+            bisect(left, left_parts, base_label)
+            bisect(right, parts - left_parts, base_label + left_parts)
+            return
+        inset = np.zeros(num_nodes, dtype=bool)
+        inset[nodes] = True
+        seed_node = nodes[0]
         chosen = np.zeros(num_nodes, dtype=bool)
         frontier = [seed_node]
         chosen[seed_node] = True
