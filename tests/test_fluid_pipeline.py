@@ -1,0 +1,122 @@
+"""Fluid113K generation pipeline cores (VERDICT round-1 gap #3).
+
+SPlisHSPlasH/partio/open3d are absent in this image; everything above
+those boundaries is exercised here: OBJ parsing + surface sampling,
+occupancy rasterization, FFT-based fluid placement, scene.json assembly
+with the reference's SPlisHSPlasH parameter blocks, and the array->record
+conversion feeding the training reader end-to-end.
+"""
+
+import json
+import os
+import sys
+
+import numpy as np
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+FLUID_DIR = os.path.join(ROOT, "dataset_generation", "Fluid113K")
+sys.path.insert(0, FLUID_DIR)
+
+import physics_data_helper as pdh  # noqa: E402
+import scene_builder as sb  # noqa: E402
+
+
+def test_obj_load_and_surface_sampling():
+    v, t = sb.load_obj(os.path.join(FLUID_DIR, "models", "Box.obj"))
+    assert v.shape == (8, 3) and t.shape == (12, 3)   # quads fan-split
+    pts, normals = sb.sample_obj_surface(
+        os.path.join(FLUID_DIR, "models", "Box.obj"))
+    assert pts.shape[0] > 1000 and pts.shape == normals.shape
+    # sampled points lie on the box surface
+    assert pts[:, 0].min() >= -2.51 and pts[:, 0].max() <= 2.51
+    on_face = (np.isclose(np.abs(pts[:, 0]), 2.5, atol=1e-4)
+               | np.isclose(np.abs(pts[:, 2]), 2.5, atol=1e-4)
+               | np.isclose(pts[:, 1], 0.0, atol=1e-4)
+               | np.isclose(pts[:, 1], 4.0, atol=1e-4))
+    assert on_face.all()
+
+
+def test_rasterize_and_place():
+    rng = np.random.default_rng(0)
+    box = sb.sample_box_volume([-1, 0, -1], [1, 2, 1], radius=0.05)
+    rast = sb.rasterize_points(box, 0.101 * 2, 0.05)
+    assert rast[2].any()
+    fluid = sb.sample_box_volume([0, 0, 0], [0.5, 0.5, 0.5], radius=0.05)
+    frast = sb.rasterize_points(fluid, 0.101 * 2, 0.05)
+    off = sb.place_fluid([rast[0], rast[1], rast[2].copy()], frast, rng)
+    placed = fluid + off
+    # placed fluid stays inside the box volume
+    assert placed.min() >= -1.3 and placed.max() <= 2.3
+
+
+def test_scene_builder_end_to_end(tmp_path):
+    info = sb.build_scene(7, os.path.join(FLUID_DIR, "models"),
+                          str(tmp_path), num_objects=2)
+    assert info["num_objects"] == 2
+    assert info["num_fluid_particles"] > 10000
+    scene = json.load(open(os.path.join(info["sim_dir"], "scene.json")))
+    # reference SPlisHSPlasH parameter surface
+    assert scene["Configuration"]["particleRadius"] == 0.025
+    assert scene["Configuration"]["simulationMethod"] == 4
+    assert scene["Simulation"]["contactTolerance"] == 0.0125
+    assert len(scene["FluidModels"]) == 2
+    assert scene["RigidBodies"][0]["isWall"] is True
+    for fm in scene["FluidModels"]:
+        assert scene[fm["id"]]["viscosity"] > 0
+    # initial states written (npz fallback without partio)
+    assert (os.path.exists(os.path.join(info["sim_dir"], "fluid0.npz"))
+            or os.path.exists(os.path.join(info["sim_dir"], "fluid0.bgeo")))
+    assert os.path.exists(os.path.join(info["sim_dir"], "box.obj"))
+
+
+def test_bgeo_discovery_helpers(tmp_path):
+    d = tmp_path / "partio"
+    d.mkdir()
+    for fid in ("fluid0", "fluid1"):
+        for i in (0, 2, 10):
+            (d / f"ParticleData_{fid}_{i}.bgeo").touch()
+    assert pdh.fluid_ids_in_dir(str(d)) == ["fluid0", "fluid1"]
+    files = pdh.fluid_bgeo_files(str(d), "fluid0")
+    assert [pdh.fluid_frame_id(f) for f in files] == [0, 2, 10]
+
+
+def test_records_roundtrip_into_training_reader(tmp_path):
+    """frames -> 16 chunk records -> distegnn reader -> preprocessing."""
+    import create_physics_records as cpr
+
+    from distegnn_amd.data.readers.fluid113k import read_fluid_sim
+
+    rng = np.random.default_rng(1)
+    N, T = 60, 32
+    frames = [{"frame_id": np.int64(i),
+               "pos": rng.random((N, 3), dtype=np.float32),
+               "vel": rng.standard_normal((N, 3)).astype(np.float32),
+               "viscosity": np.full(N, 0.02, dtype=np.float32),
+               "m": np.full(N, 0.125, dtype=np.float32)}
+              for i in range(T)]
+    chunks = cpr.frames_to_chunks(frames, 16)
+    assert len(chunks) == 16 and sum(len(c) for c in chunks) == T
+    paths = cpr.write_records(frames, str(tmp_path / "sim_0001"),
+                              compress=False)
+    assert len(paths) == 16
+    pos, vel, visc, mass = read_fluid_sim(str(tmp_path), 1)
+    assert pos.shape == (T, N, 3)
+    assert np.allclose(pos[5], frames[5]["pos"])
+    assert float(visc[0]) == pytest.approx(0.02)
+
+
+def test_records_zstd_guard(tmp_path):
+    """.zst chunks raise a clear ImportError without zstandard."""
+    try:
+        import zstandard  # noqa: F401
+
+        pytest.skip("zstandard installed; guard not reachable")
+    except ImportError:
+        pass
+    from distegnn_amd.data.readers.fluid113k import _read_chunk
+
+    p = tmp_path / "sim_0001_00.msgpack.zst"
+    p.write_bytes(b"\x28\xb5\x2f\xfd")        # zstd magic
+    with pytest.raises(ImportError, match="zstandard"):
+        _read_chunk(str(p))
