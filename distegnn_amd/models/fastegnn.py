@@ -236,9 +236,10 @@ class EGCLVel(nn.Module):
                 agg = ops.segment_sum(trans, row, n, rowptr=rowptr)
             else:
                 raise ValueError(f"coords_agg {self.coords_agg}")
-        coord = coord + agg
-        coord = coord + trans_v
-        coord = coord + self.coord_mlp_vel(h) * vel
+        # fused tail: coord + agg + trans_v + phi_v(h) * vel in one
+        # kernel pair (ops.coord_update; eager composition off-GPU)
+        coord = ops.coord_update(coord, agg, trans_v,
+                                 self.coord_mlp_vel(h), vel)
         if self.gravity is not None:
             coord = coord + self.gravity_mlp(h) * self.gravity.to(h.device)
 

@@ -192,6 +192,42 @@ def mid_mean(x: torch.Tensor) -> torch.Tensor:
     return x.mean(dim=1)
 
 
+class _CoordUpdateFn(torch.autograd.Function):
+    """Fused coordinate-update tail: out = coord + agg + trans_v +
+    phi_v * vel (one kernel each way; reference FastEGNN.py:166-188 runs
+    this as 3 adds + broadcast mul + mirrored backward, ~8 launches)."""
+
+    @staticmethod
+    def forward(ctx, coord, agg, trans_v, phiv, vel):
+        ext = _require_ext("coord_update")
+        if ext is None:
+            ctx.save_for_backward(vel)
+            return coord + agg + trans_v + phiv * vel
+        ctx.save_for_backward(vel)
+        return ext.coord_update_forward(coord, agg, trans_v, phiv, vel)
+
+    @staticmethod
+    def backward(ctx, g):
+        (vel,) = ctx.saved_tensors
+        g = g.contiguous()
+        ext = _load_extension()
+        if ext is None or not g.is_cuda:
+            dphiv = (g * vel).sum(-1, keepdim=True)
+        else:
+            dphiv = ext.coord_update_backward(g, vel)
+        return g, g, g, dphiv, None
+
+
+def coord_update(coord, agg, trans_v, phiv, vel):
+    """coord + agg + trans_v + phi_v * vel, HIP-fused on GPU fp32."""
+    if (coord.is_cuda and coord.dtype == torch.float32
+            and hip_ext() is not None
+            and hasattr(hip_ext(), "coord_update_forward")):
+        return _CoordUpdateFn.apply(coord, agg.float(), trans_v.float(),
+                                    phiv.float(), vel)
+    return coord + agg + trans_v + phiv * vel
+
+
 class _GatherRowsFn(torch.autograd.Function):
     """index_select(0, idx) whose BACKWARD is a deterministic CSR segment
     sum instead of torch's index_add scatter.
@@ -622,5 +658,5 @@ __all__ = [
     "segment_sum", "segment_mean", "graph_sum_pool", "graph_mean_pool",
     "gather_rows", "fused_edge_block", "eager_edge_block",
     "fused_virtual_block", "eager_virtual_block", "radius_graph",
-    "hip_ext", "reference", "refresh_weight_prep",
+    "hip_ext", "reference", "refresh_weight_prep", "coord_update",
 ]

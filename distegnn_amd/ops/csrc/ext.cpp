@@ -22,6 +22,10 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     torch::Tensor w3v, bool normalize, double eps,
     std::vector<torch::Tensor> prepped = {});
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt);
+torch::Tensor coord_update_forward(torch::Tensor coord, torch::Tensor agg,
+                                   torch::Tensor trans_v,
+                                   torch::Tensor phiv, torch::Tensor vel);
+torch::Tensor coord_update_backward(torch::Tensor g, torch::Tensor vel);
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
 torch::Tensor tall_linear(torch::Tensor x, torch::Tensor bmat,
                           c10::optional<torch::Tensor> bias, int64_t act);
@@ -120,6 +124,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused MFMA virtual-edge block forward");
   m.def("fused_virtual_backward", &fused_virtual_backward,
         "fused MFMA virtual-edge block backward");
+  m.def("coord_update_forward", &coord_update_forward,
+        "fused coord + agg + trans_v + phi_v*vel");
+  m.def("coord_update_backward", &coord_update_backward,
+        "dphiv = sum_d g*vel for the fused coordinate update");
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
         py::arg("a"), py::arg("bt"));

@@ -171,10 +171,19 @@ class _FusedAllReduceSum(torch.autograd.Function):
 
 def _drain_watchdog():
     """Let the ProcessGroupNCCL watchdog dequeue outstanding works before a
-    capture (its hipEventQuery during capture aborts the process)."""
+    capture (its hipEventQuery during capture aborts the process).
+
+    LOCAL drain only — deliberately NO dist.barrier(): the watchdog
+    watches this rank's works, which a device synchronize completes in a
+    lockstep schedule (peers have issued the matching collectives by
+    construction). A default-group barrier here can deadlock when ranks
+    reach their capture points at different step indices (per-rank shape
+    keys: one rank's batches may collide in shape and hit the
+    warmup-occurrence threshold earlier than its peers, who are busy
+    issuing capture-communicator collectives and never touch the default
+    group inside the replay window)."""
     import time as _time
 
-    dist.barrier()
     torch.cuda.synchronize()
     _time.sleep(0.5)
 

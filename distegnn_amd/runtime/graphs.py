@@ -217,13 +217,15 @@ class GraphedStep:
                     and torch.distributed.is_initialized()):
                 # Drain the ProcessGroupNCCL watchdog before capturing:
                 # the watchdog thread polls hipEventQuery on outstanding
-                # pre-capture collectives (e.g. the per-step counts_global
-                # all-reduce) and an event query DURING stream capture is
-                # hipErrorStreamCaptureUnsupported -> process abort.
-                # Collectives issued INSIDE the capture are not watched.
+                # pre-capture collectives and an event query DURING stream
+                # capture is hipErrorStreamCaptureUnsupported -> process
+                # abort. Collectives issued INSIDE the capture are not
+                # watched. LOCAL drain only — a default-group barrier here
+                # deadlocks when shape-key collisions make ranks reach
+                # their capture points at different step indices (see
+                # parallel/comm._drain_watchdog).
                 import time as _time
 
-                torch.distributed.barrier()
                 torch.cuda.synchronize()
                 _time.sleep(0.5)
             g = torch.cuda.CUDAGraph()
