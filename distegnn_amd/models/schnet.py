@@ -11,6 +11,7 @@ Module names match for checkpoint compatibility.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 import torch.nn.functional as F
@@ -63,10 +64,26 @@ class CFConv(nn.Module):
 
     def forward(self, x, edge_index, edge_weight, edge_attr, rowptr=None,
                 colptr=None, col_perm=None):
-        c = 0.5 * (torch.cos(edge_weight * math.pi / self.cutoff) + 1.0)
-        w = self.nn(edge_attr) * c.view(-1, 1)
         x = self.lin1(x)
-        msg = ops.gather_rows(x, edge_index[1], colptr, col_perm) * w
+        sm = getattr(self, "smearing", None)
+        fused = (sm is not None and x.is_cuda
+                 and x.dtype == torch.bfloat16 and x.size(1) in (64, 128)
+                 and sm.offset.numel() <= 64 and rowptr is not None
+                 and colptr is not None and ops.hip_ext() is not None
+                 and hasattr(ops.hip_ext(), "cfconv_forward")
+                 and os.environ.get("DISTEGNN_DISABLE_FUSED") != "1")
+        if fused:
+            # one kernel: smearing + filter MLP + cosine cutoff + gathered
+            # multiply (csrc/cfconv.hip)
+            msg = ops.cfconv_msg(
+                x.contiguous(), edge_weight, edge_index[0], edge_index[1],
+                colptr, col_perm, self.nn[0].weight, self.nn[0].bias,
+                self.nn[2].weight, self.nn[2].bias, sm.offset, sm.coeff,
+                self.cutoff)
+        else:
+            c = 0.5 * (torch.cos(edge_weight * math.pi / self.cutoff) + 1.0)
+            w = self.nn(edge_attr) * c.view(-1, 1)
+            msg = ops.gather_rows(x, edge_index[1], colptr, col_perm) * w
         agg = ops.segment_sum(msg, edge_index[0], x.size(0), rowptr=rowptr)
         return self.lin2(agg)
 
@@ -139,6 +156,10 @@ class SchNet(nn.Module):
         self.lin1 = nn.Linear(hidden_channels, hidden_channels // 2)
         self.act = ShiftedSoftplus()
         self.lin2 = nn.Linear(hidden_channels // 2, 1)
+        for ib in self.interactions:
+            # non-module attribute: lets CFConv fuse the smearing into its
+            # HIP kernel without re-owning the buffer (state-dict unchanged)
+            object.__setattr__(ib.conv, "smearing", self.distance_expansion)
 
     def forward(self, z, pos, edge_index, batch=None, embedding=True, *,
                 rowptr=None, colptr=None, col_perm=None, **unused):

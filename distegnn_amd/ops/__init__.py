@@ -228,6 +228,59 @@ def coord_update(coord, agg, trans_v, phiv, vel):
     return coord + agg + trans_v + phiv * vel
 
 
+class _CFConvMsgFn(torch.autograd.Function):
+    """Fused SchNet CFConv messages (csrc/cfconv.hip): smearing + filter
+    MLP + cosine cutoff + gathered multiply in one kernel. Backward
+    recomputes the eager composition under autograd (its gathers use the
+    CSR segment-sum backward — no index_add scatters)."""
+
+    @staticmethod
+    def forward(ctx, xw1, dist, row, col, colptr, col_perm,
+                w1, b1, w2, b2, offsets, coeff, cutoff):
+        ext = _require_ext("cfconv")
+        from . import prep
+
+        msg = ext.cfconv_forward(
+            xw1, dist, row, col, prep.get(w1, "pad_gpad"), b1,
+            prep.get(w2, "bf16"), b2, offsets, float(coeff), float(cutoff))
+        ctx.save_for_backward(xw1, dist, row, col, colptr, col_perm,
+                              w1, b1, w2, b2, offsets)
+        ctx.coeff, ctx.cutoff = coeff, cutoff
+        return msg
+
+    @staticmethod
+    def backward(ctx, gmsg):
+        import math
+
+        import torch.nn.functional as F
+
+        (xw1, dist, row, col, colptr, col_perm,
+         w1, b1, w2, b2, offsets) = ctx.saved_tensors
+        with torch.enable_grad():
+            leaves = [t.detach().requires_grad_(True)
+                      for t in (xw1, w1, b1, w2, b2)]
+            lx, lw1, lb1, lw2, lb2 = leaves
+            d = dist.view(-1, 1) - offsets.view(1, -1)
+            gauss = torch.exp(ctx.coeff * d.pow(2)).to(lx.dtype)
+            c = 0.5 * (torch.cos(dist * math.pi / ctx.cutoff) + 1.0)
+            z1 = F.softplus(F.linear(gauss, lw1, lb1)) - math.log(2.0)
+            w = F.linear(z1.to(lx.dtype), lw2, lb2)
+            xj = gather_rows(lx, col, colptr, col_perm)
+            msg = xj * w * c.view(-1, 1).to(lx.dtype)
+            grads = torch.autograd.grad(
+                msg, leaves, grad_outputs=gmsg.to(msg.dtype),
+                allow_unused=True)
+        return (grads[0], None, None, None, None, None, grads[1],
+                grads[2], grads[3], grads[4], None, None, None)
+
+
+def cfconv_msg(xw1, dist, row, col, colptr, col_perm, w1, b1, w2, b2,
+               offsets, coeff, cutoff):
+    """Per-edge CFConv messages; aggregate with segment_sum (K14)."""
+    return _CFConvMsgFn.apply(xw1, dist, row, col, colptr, col_perm,
+                              w1, b1, w2, b2, offsets, coeff, cutoff)
+
+
 class _GatherRowsFn(torch.autograd.Function):
     """index_select(0, idx) whose BACKWARD is a deterministic CSR segment
     sum instead of torch's index_add scatter.
@@ -659,4 +712,5 @@ __all__ = [
     "gather_rows", "fused_edge_block", "eager_edge_block",
     "fused_virtual_block", "eager_virtual_block", "radius_graph",
     "hip_ext", "reference", "refresh_weight_prep", "coord_update",
+    "cfconv_msg",
 ]

@@ -27,6 +27,7 @@ SOURCES = [
     os.path.join(_CSRC, "tall_linear.hip"),
     os.path.join(_CSRC, "fused_virtual.hip"),
     os.path.join(_CSRC, "elementwise.hip"),
+    os.path.join(_CSRC, "cfconv.hip"),
 ]
 
 

@@ -26,6 +26,12 @@ torch::Tensor coord_update_forward(torch::Tensor coord, torch::Tensor agg,
                                    torch::Tensor trans_v,
                                    torch::Tensor phiv, torch::Tensor vel);
 torch::Tensor coord_update_backward(torch::Tensor g, torch::Tensor vel);
+torch::Tensor cfconv_forward(torch::Tensor xw1, torch::Tensor dist,
+                             torch::Tensor row, torch::Tensor col,
+                             torch::Tensor w1f, torch::Tensor b1,
+                             torch::Tensor w2f, torch::Tensor b2,
+                             torch::Tensor offsets, double coeff,
+                             double cutoff);
 torch::Tensor wgrad_splitk_launch(torch::Tensor g, torch::Tensor x);
 torch::Tensor tall_linear(torch::Tensor x, torch::Tensor bmat,
                           c10::optional<torch::Tensor> bias, int64_t act);
@@ -128,6 +134,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused coord + agg + trans_v + phi_v*vel");
   m.def("coord_update_backward", &coord_update_backward,
         "dphiv = sum_d g*vel for the fused coordinate update");
+  m.def("cfconv_forward", &cfconv_forward,
+        "fused SchNet CFConv messages: smearing + filter MLP + cutoff + "
+        "gathered multiply");
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA layout probe: D = A @ B (bt = B^T)",
         py::arg("a"), py::arg("bt"));

@@ -55,9 +55,14 @@ def t_bf16(t: torch.Tensor) -> torch.Tensor:
     return t.detach().bfloat16().t().contiguous()
 
 
+def pad_gpad(t: torch.Tensor) -> torch.Tensor:
+    """[F, G] -> [F, 64] bf16 (CFConv filter L1, gaussian dim padded)."""
+    return F.pad(t.detach().bfloat16(), (0, 64 - t.size(1))).contiguous()
+
+
 _TRANSFORMS: Dict[str, Callable[[torch.Tensor], torch.Tensor]] = {
     "bf16": bf16c, "f32": f32c, "pad_kpad": pad_kpad,
-    "tpad_kout": tpad_kout, "t_bf16": t_bf16,
+    "tpad_kout": tpad_kout, "t_bf16": t_bf16, "pad_gpad": pad_gpad,
 }
 
 # (id(tensor), kind) -> [version, buffer, source-tensor-ref]

@@ -238,3 +238,64 @@ def test_coord_update_matches_eager():
     out_e.backward(g)
     for a, b in zip(leaves_f, leaves_e):
         assert torch.allclose(a.grad, b.grad, atol=1e-6)
+
+
+@pytest.mark.parametrize("f", [64, 128])
+def test_cfconv_fused_matches_eager(f):
+    """Fused CFConv messages (smearing+filter MLP+cutoff+gather mul) ==
+    the eager composition, fwd + bwd."""
+    import math
+
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    dev_ = "cuda:0"
+    n, m, g = 500, 4000, 50
+    cutoff = 10.0
+    x = (torch.randn(n, f, device=dev_) * 0.5).bfloat16()
+    row = torch.sort(torch.randint(0, n, (m,), device=dev_)).values
+    col = torch.randint(0, n, (m,), device=dev_)
+    dist = torch.rand(m, device=dev_) * cutoff
+    # CSR metadata
+    rowptr = torch.zeros(n + 1, dtype=torch.long, device=dev_)
+    rowptr.scatter_add_(0, row + 1, torch.ones_like(row))
+    rowptr = rowptr.cumsum(0)
+    col_perm = torch.argsort(col, stable=True)
+    colptr = torch.zeros(n + 1, dtype=torch.long, device=dev_)
+    colptr.scatter_add_(0, col[col_perm] + 1, torch.ones_like(col))
+    colptr = colptr.cumsum(0)
+    offsets = torch.linspace(0.0, cutoff, g, device=dev_)
+    coeff = -0.5 / float(offsets[1] - offsets[0]) ** 2
+    w1 = torch.randn(f, g, device=dev_) * 0.2
+    b1 = torch.randn(f, device=dev_) * 0.1
+    w2 = torch.randn(f, f, device=dev_) * 0.1
+    b2 = torch.randn(f, device=dev_) * 0.1
+
+    def eager(leaves):
+        lx, lw1, lb1, lw2, lb2 = leaves
+        d = dist.view(-1, 1) - offsets.view(1, -1)
+        gauss = torch.exp(coeff * d.pow(2)).bfloat16()
+        c = 0.5 * (torch.cos(dist * math.pi / cutoff) + 1.0)
+        z1 = (F.softplus(F.linear(gauss, lw1.bfloat16(), lb1.bfloat16()))
+              - math.log(2.0))
+        w = F.linear(z1.bfloat16(), lw2.bfloat16(), lb2.bfloat16())
+        xj = ops.gather_rows(lx, col, colptr, col_perm)
+        return xj * w * c.view(-1, 1).bfloat16()
+
+    leaves_e = [t.clone().requires_grad_(True) for t in (x, w1, b1, w2, b2)]
+    msg_e = eager(leaves_e)
+    leaves_f = [t.clone().requires_grad_(True) for t in (x, w1, b1, w2, b2)]
+    msg_f = ops.cfconv_msg(leaves_f[0], dist, row, col, colptr, col_perm,
+                           leaves_f[1], leaves_f[2], leaves_f[3],
+                           leaves_f[4], offsets, coeff, cutoff)
+    assert msg_f.shape == (m, f)
+    assert torch.allclose(msg_f.float(), msg_e.float(), atol=0.05,
+                          rtol=0.05), (msg_f.float() - msg_e.float()
+                                       ).abs().max()
+    gout = torch.randn_like(msg_e)
+    msg_e.backward(gout)
+    msg_f.backward(gout)
+    for i, (a, b) in enumerate(zip(leaves_f, leaves_e)):
+        assert torch.allclose(a.grad.float(), b.grad.float(), atol=0.1,
+                              rtol=0.1), (i, (a.grad.float()
+                                              - b.grad.float()).abs().max())
