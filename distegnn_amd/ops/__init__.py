@@ -263,8 +263,12 @@ class _CFConvMsgFn(torch.autograd.Function):
             d = dist.view(-1, 1) - offsets.view(1, -1)
             gauss = torch.exp(ctx.coeff * d.pow(2)).to(lx.dtype)
             c = 0.5 * (torch.cos(dist * math.pi / ctx.cutoff) + 1.0)
-            z1 = F.softplus(F.linear(gauss, lw1, lb1)) - math.log(2.0)
-            w = F.linear(z1.to(lx.dtype), lw2, lb2)
+            # leaves stay in parameter dtype (fp32); grads flow through
+            # the compute-dtype casts
+            z1 = F.softplus(F.linear(gauss, lw1.to(gauss.dtype),
+                                     lb1.to(gauss.dtype))) - math.log(2.0)
+            w = F.linear(z1.to(lx.dtype), lw2.to(lx.dtype),
+                         lb2.to(lx.dtype))
             xj = gather_rows(lx, col, colptr, col_perm)
             msg = xj * w * c.view(-1, 1).to(lx.dtype)
             grads = torch.autograd.grad(
