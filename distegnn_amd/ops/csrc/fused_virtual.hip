@@ -115,6 +115,39 @@ __host__ __device__ constexpr VSmem vsmem_layout() {
   return L;
 }
 
+// Backward-only layout: the staging region (sa) is H_STRIDE-wide — the
+// dvin product streams through it in three 48-column passes instead of
+// needing a K_STRIDE-wide tile — and the bias slot holds only wxvv/wXv.
+// Total is exactly 40960 B/block -> 4 blocks/CU (the 122-VGPR backward
+// allows 4 waves/SIMD; the full-width forward layout caps both at 3).
+struct VSmemB {
+  int sa;     // [TILE][H_STRIDE] bf16 (zX staging -> dzX -> z2 staging ->
+              //   dvin pass window)
+  int za;     // [TILE][H_STRIDE] bf16
+  int zb;     // [TILE][H_STRIDE] bf16
+  int zc;     // [TILE][H_STRIDE] bf16
+  int diff;   // [TILE][4] f32
+  int scal;   // [TILE][4] f32
+  int bias;   // [2*H] f32 (wxvv | wXv)
+  int gbacc;  // [6*H] f32
+  int total;
+};
+
+__host__ __device__ constexpr VSmemB vsmem_layout_bwd() {
+  VSmemB L{};
+  int o = 0;
+  L.sa = o; o += TILE * H_STRIDE * 2;
+  L.za = o; o += TILE * H_STRIDE * 2;
+  L.zb = o; o += TILE * H_STRIDE * 2;
+  L.zc = o; o += TILE * H_STRIDE * 2;
+  L.diff = o; o += TILE * 4 * 4;
+  L.scal = o; o += TILE * 4 * 4;
+  L.bias = o; o += 2 * H * 4;
+  L.gbacc = o; o += 6 * H * 4;
+  L.total = o;
+  return L;
+}
+
 // C-layout epilogue write into an LDS bf16 tile
 #define EPI_WRITE(tileptr, expr)                                            \
   _Pragma("unroll") for (int nt = 0; nt < 4; ++nt) {                        \
@@ -321,7 +354,7 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
   }
 }
 
-__global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
+__global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
     const float* __restrict__ coord, const float* __restrict__ vcoord,
     const long* __restrict__ batch,
     const bf16* __restrict__ dvmsg,    // [R,64] cotangent of vmsg
@@ -344,16 +377,16 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
     float* __restrict__ gb_out,  // [6H]: gb1|gb2|gbxv|gbX|gwxvv|gwXv
     long n_rows, int cdim, int k_in) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr VSmem L = vsmem_layout();
+  constexpr VSmemB LB = vsmem_layout_bwd();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  float* biases = reinterpret_cast<float*>(smem + L.bias);
-  float* gbacc = reinterpret_cast<float*>(smem + L.gbacc);
+  float* biases = reinterpret_cast<float*>(smem + LB.bias);
+  float* gbacc = reinterpret_cast<float*>(smem + LB.gbacc);
   for (int i = tid; i < H; i += THREADS) {
-    biases[4 * H + i] = wxvv[i];
-    biases[5 * H + i] = wXv[i];
+    biases[i] = wxvv[i];
+    biases[H + i] = wXv[i];
     for (int kacc = 0; kacc < 6; ++kacc) gbacc[kacc * H + i] = 0.f;
   }
 
@@ -365,8 +398,8 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
 
     // per-row: vdiff recompute, dpxv/dpX, initial dvd
     for (int e = tid; e < TILE; e += THREADS) {
-      float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
-      float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
+      float* dptr = reinterpret_cast<float*>(smem + LB.diff) + e * 4;
+      float* sc = reinterpret_cast<float*>(smem + LB.scal) + e * 4;
       float dpxv = 0, dpX = 0;
       if (e < nrow) {
         long r = r0 + e;
@@ -401,21 +434,21 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
         v = g8(zxv_in + (r0 + e) * H + c8);
         w = g8(zX_in + (r0 + e) * H + c8);
       }
-      *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
-      *reinterpret_cast<bf16x8*>(smem + L.in_tile
+      *reinterpret_cast<bf16x8*>(smem + LB.za + (e * H_STRIDE + c8) * 2) = v;
+      *reinterpret_cast<bf16x8*>(smem + LB.sa
                                  + (e * H_STRIDE + c8) * 2) = w;
     }
     __syncthreads();
     {
       int cc = tid & 63;
       int estart = (tid >> 6) * 16;
-      float wv0 = biases[4 * H + cc], wv1 = biases[5 * H + cc];
-      __bf16* zc = reinterpret_cast<__bf16*>(smem + L.zc);
+      float wv0 = biases[cc], wv1 = biases[H + cc];
+      __bf16* zc = reinterpret_cast<__bf16*>(smem + LB.zc);
       // dzX overwrites the zX staging in place (same slot, same thread)
-      __bf16* zd = reinterpret_cast<__bf16*>(smem + L.in_tile);
-      const __bf16* zxs = reinterpret_cast<const __bf16*>(smem + L.za);
+      __bf16* zd = reinterpret_cast<__bf16*>(smem + LB.sa);
+      const __bf16* zxs = reinterpret_cast<const __bf16*>(smem + LB.za);
       const __bf16* zXs = zd;
-      const float* sc = reinterpret_cast<const float*>(smem + L.scal);
+      const float* sc = reinterpret_cast<const float*>(smem + LB.scal);
       float abxv = 0.f, abX = 0.f, awxvv = 0.f, awXv = 0.f;
       for (int e = estart; e < estart + 16; ++e) {
         bool ok = e < nrow;
@@ -439,15 +472,15 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       atomicAdd(&gbacc[5 * H + cc], awXv);
     }
     __syncthreads();
-    TILE_TO_GLOBAL(L.zc, dzxv_out, false);
-    TILE_TO_GLOBAL(L.in_tile, dzX_out, false);
+    TILE_TO_GLOBAL(LB.zc, dzxv_out, false);
+    TILE_TO_GLOBAL(LB.sa, dzX_out, false);
 
     // dvmsg_tot = dvmsg + dzxv@Wxv + dzX@WX; dz2 = dvmsg_tot o silu'(z2)
     {
       f32x4 acc[4] = {};
-      mm_g<2, false>(smem, L.zc + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      mm_g<2, false>(smem, LB.zc + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(wxvt), H, lane, acc);
-      mm_g<2, false>(smem, L.in_tile + wave * 16 * H_STRIDE * 2,
+      mm_g<2, false>(smem, LB.sa + wave * 16 * H_STRIDE * 2,
                      H_STRIDE * 2, opaque(wXt), H, lane, acc);
       __syncthreads();
       // in_tile consumed: re-stage za/in_tile with dvmsg and z2, coalesced
@@ -459,14 +492,14 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
           v = g8(dvmsg + (r0 + e) * H + c8);
           w = g8(z2_in + (r0 + e) * H + c8);
         }
-        *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
-        *reinterpret_cast<bf16x8*>(smem + L.in_tile
+        *reinterpret_cast<bf16x8*>(smem + LB.za + (e * H_STRIDE + c8) * 2) = v;
+        *reinterpret_cast<bf16x8*>(smem + LB.sa
                                    + (e * H_STRIDE + c8) * 2) = w;
       }
       __syncthreads();
-      __bf16* zb = reinterpret_cast<__bf16*>(smem + L.zb);
-      const __bf16* ups = reinterpret_cast<const __bf16*>(smem + L.za);
-      const __bf16* z2s = reinterpret_cast<const __bf16*>(smem + L.in_tile);
+      __bf16* zb = reinterpret_cast<__bf16*>(smem + LB.zb);
+      const __bf16* ups = reinterpret_cast<const __bf16*>(smem + LB.za);
+      const __bf16* z2s = reinterpret_cast<const __bf16*>(smem + LB.sa);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int cc = nt * 16 + (lane & 15);
@@ -481,11 +514,11 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       }
     }
     __syncthreads();
-    TILE_TO_GLOBAL(L.zb, dz2_out, false);
+    TILE_TO_GLOBAL(LB.zb, dz2_out, false);
     {
       int cc = tid & 63;
       int estart = (tid >> 6) * 16;
-      const __bf16* zb = reinterpret_cast<const __bf16*>(smem + L.zb);
+      const __bf16* zb = reinterpret_cast<const __bf16*>(smem + LB.zb);
       float ab = 0.f;
       for (int e = estart; e < estart + 16; ++e)
         ab += (float)zb[e * H_STRIDE + cc];
@@ -500,13 +533,13 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
         int c8 = (idx % 8) * 8;
         bf16x8 v = {};
         if (e < nrow) v = g8(z1_in + (r0 + e) * H + c8);
-        *reinterpret_cast<bf16x8*>(smem + L.za + (e * H_STRIDE + c8) * 2) = v;
+        *reinterpret_cast<bf16x8*>(smem + LB.za + (e * H_STRIDE + c8) * 2) = v;
       }
       f32x4 acc[4] = {};
-      mm_g<2, false>(smem, L.zb + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      mm_g<2, false>(smem, LB.zb + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w2t), H, lane, acc);
       __syncthreads();
-      __bf16* za = reinterpret_cast<__bf16*>(smem + L.za);
+      __bf16* za = reinterpret_cast<__bf16*>(smem + LB.za);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int cc = nt * 16 + (lane & 15);
@@ -519,18 +552,22 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
       }
     }
     __syncthreads();
-    TILE_TO_GLOBAL(L.za, dz1_out, false);
+    TILE_TO_GLOBAL(LB.za, dz1_out, false);
     {
       int cc = tid & 63;
       int estart = (tid >> 6) * 16;
-      const __bf16* za = reinterpret_cast<const __bf16*>(smem + L.za);
+      const __bf16* za = reinterpret_cast<const __bf16*>(smem + LB.za);
       float ab = 0.f;
       for (int e = estart; e < estart + 16; ++e)
         ab += (float)za[e * H_STRIDE + cc];
       atomicAdd(&gbacc[cc], ab);
     }
 
-    // dvin = dz1 @ W1 (3 register passes x 3 n-tiles), into in_tile
+    // dvin = dz1 @ W1, streamed through the H_STRIDE-wide sa window in
+    // three 48-column passes (a K_STRIDE-wide tile would push the block
+    // past the 40 KB that fits 4 blocks/CU). Logical column cc lives at
+    // physical cc - pass*48; each pass's dh/dvf columns are stored before
+    // the next pass overwrites the window.
 #pragma unroll
     for (int pass = 0; pass < 3; ++pass) {
       const bf16* w1tp_ = opaque(w1tp);
@@ -538,7 +575,7 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
         int k = kk * 32 + (lane >> 4) * 8;
-        bf16x8 a = lds8(smem, L.za + ((wave * 16 + (lane & 15)) * H_STRIDE
+        bf16x8 a = lds8(smem, LB.za + ((wave * 16 + (lane & 15)) * H_STRIDE
                                       + k) * 2);
 #pragma unroll
         for (int nt = 0; nt < 3; ++nt) {
@@ -548,44 +585,50 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_bwd(
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
         }
       }
-      __bf16* dvin = reinterpret_cast<__bf16*>(smem + L.in_tile);
+      __syncthreads();  // prior pass's sa readers have drained
+      __bf16* dvin = reinterpret_cast<__bf16*>(smem + LB.sa);
 #pragma unroll
       for (int nt = 0; nt < 3; ++nt) {
         int cc = (pass * 3 + nt) * 16 + (lane & 15);
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) {
           int e = wave * 16 + (lane >> 4) * 4 + rr;
-          dvin[e * K_STRIDE + cc] = (__bf16)acc[nt][rr];
+          dvin[e * H_STRIDE + cc - pass * 48] = (__bf16)acc[nt][rr];
           if (cc == 2 * H) {
-            reinterpret_cast<float*>(smem + L.scal)[e * 4 + 2] =
+            reinterpret_cast<float*>(smem + LB.scal)[e * 4 + 2] =
                 acc[nt][rr];  // d(vrad)
           }
         }
       }
+      __syncthreads();
+      // this pass's dh/dvf columns -> global (48-col windows stay
+      // 8-aligned; cols >= 2H are the vrad/dgram tail handled below)
+      for (int idx = tid; idx < TILE * 6; idx += THREADS) {
+        int e = idx / 6;
+        if (e >= nrow) continue;
+        int c8 = pass * 48 + (idx % 6) * 8;
+        if (c8 >= 2 * H) continue;
+        bf16x8 v = lds8(smem, LB.sa + (e * H_STRIDE + c8 - pass * 48) * 2);
+        if (c8 < H)
+          *reinterpret_cast<bf16x8*>(dh_out + (r0 + e) * H + c8) = v;
+        else
+          *reinterpret_cast<bf16x8*>(dvf_out + (r0 + e) * H + c8 - H) = v;
+      }
     }
-    __syncthreads();
 
-    // dh / dvf (coalesced), dgram + dvd (per-row)
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
-      if (e >= nrow) continue;
-      int c8 = (idx % 8) * 8;
-      *reinterpret_cast<bf16x8*>(dh_out + (r0 + e) * H + c8) =
-          lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
-      *reinterpret_cast<bf16x8*>(dvf_out + (r0 + e) * H + c8) =
-          lds8(smem, L.in_tile + (e * K_STRIDE + H + c8) * 2);
-    }
+    // dgram (logical cols 2H+1.. live in the pass-2 window at physical
+    // offset -96) + dvd (per-row)
     for (int e = tid; e < nrow; e += THREADS) {
       long r = r0 + e;
       const __bf16* dvin = reinterpret_cast<const __bf16*>(
-          smem + L.in_tile) + e * K_STRIDE;
+          smem + LB.sa) + e * H_STRIDE;
       for (int j = 0; j < cdim; ++j)
-        dgram_out[r * CMAX + j] = (float)dvin[2 * H + 1 + j];
+        dgram_out[r * CMAX + j] = (float)dvin[2 * H + 1 + j - 96];
       for (int j = cdim; j < CMAX; ++j) dgram_out[r * CMAX + j] = 0.f;
       const float* dptr = reinterpret_cast<const float*>(
-          smem + L.diff) + e * 4;
+          smem + LB.diff) + e * 4;
       const float* sc = reinterpret_cast<const float*>(
-          smem + L.scal) + e * 4;
+          smem + LB.scal) + e * 4;
       float p_xv = p2_in[r * 2], p_x = p2_in[r * 2 + 1];
       float drad = sc[2];
       float inv = dptr[3] > 0.f ? drad / dptr[3] : 0.f;
@@ -714,7 +757,7 @@ std::vector<torch::Tensor> fused_virtual_backward(
   if (rows == 0)
     return {dz1, dz2, dzxv, dzX, dh, dvf, dgram, dvd, dp2, gb};
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr VSmem L = vsmem_layout();
+  constexpr VSmemB L = vsmem_layout_bwd();
   long tiles = (rows + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
   torch::Tensor w1tp, w2tc, wxvtc, wXtc, wxvvc, wXvc;
@@ -735,7 +778,7 @@ std::vector<torch::Tensor> fused_virtual_backward(
   }
   auto cc_ = coord.contiguous().to(torch::kFloat);
   auto vc = vcoord.contiguous().to(torch::kFloat);
-  fused_virtual_bwd<<<blocks, THREADS, L.zd, stream>>>(
+  fused_virtual_bwd<<<blocks, THREADS, L.total, stream>>>(
       cc_.data_ptr<float>(), vc.data_ptr<float>(),
       batch.contiguous().data_ptr<long>(),
       reinterpret_cast<const bf16*>(dvmsg.contiguous().data_ptr()),
