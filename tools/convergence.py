@@ -30,20 +30,30 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 
-def make_config(tmp, epochs):
+def make_config(tmp, epochs, workload="nbody"):
     from distegnn_amd.utils import AttrDict
 
+    if workload == "water3d":
+        data = {"data_dir": os.path.join(tmp, "data"),
+                "dataset_name": "Water-3D", "max_samples": 3000,
+                "batch_size": 4, "delta_t": 20,
+                "accelerate_mode": "cutoff_edges", "radius": 0.035,
+                "cutoff_rate": 0.0, "synthetic": True,
+                "synthetic_samples": 60, "num_workers": 0,
+                "world_size": 1}
+    else:
+        data = {"data_dir": os.path.join(tmp, "data"),
+                "dataset_name": "nbody_100", "max_samples": 3000,
+                "batch_size": 20, "frame_0": 30, "frame_T": 40,
+                "accelerate_mode": "cutoff_edges", "radius": -1,
+                "cutoff_rate": 0.0, "synthetic": True,
+                "synthetic_samples": 200, "num_workers": 0,
+                "world_size": 1}
     return AttrDict({
-        "model": {"model_name": "FastEGNN", "normalize": True,
+        "model": {"model_name": "FastEGNN", "normalize": workload == "nbody",
                   "hidden_nf": 64, "n_layers": 4, "virtual_channels": 3,
                   "node_feat_nf": 2, "node_attr_nf": 0, "edge_attr_nf": 2},
-        "data": {"data_dir": os.path.join(tmp, "data"),
-                 "dataset_name": "nbody_100", "max_samples": 3000,
-                 "batch_size": 20, "frame_0": 30, "frame_T": 40,
-                 "accelerate_mode": "cutoff_edges", "radius": -1,
-                 "cutoff_rate": 0.0, "synthetic": True,
-                 "synthetic_samples": 200, "num_workers": 0,
-                 "world_size": 1},
+        "data": data,
         "train": {"learning_rate": 5e-4, "weight_decay": 1e-12,
                   "epochs": epochs, "early_stop": 10000,
                   "mmd": {"sigma": 1.5, "weight": 0.03, "samples": 3},
@@ -105,14 +115,16 @@ def main():
     ap.add_argument("--out", type=str,
                     default="profiles/convergence_nbody.json")
     ap.add_argument("--tmp", type=str, default="/tmp/convergence")
+    ap.add_argument("--workload", type=str, default="nbody",
+                    choices=["nbody", "water3d"])
     args = ap.parse_args()
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
 
-    results = {"config": "nbody_100 FastEGNN H=64 L=4 C=3, 200 synthetic "
-                         "samples, batch 20, Adam 5e-4, seed 43",
+    results = {"config": f"{args.workload} FastEGNN H=64 L=4 C=3, "
+                         f"synthetic, Adam 5e-4, seed 43",
                "device": str(device)}
     for mode in ("reference", "fast"):
-        cfg = make_config(args.tmp, args.epochs)
+        cfg = make_config(args.tmp, args.epochs, args.workload)
         print(f"=== {mode} run ===", flush=True)
         results[mode] = run_once(cfg, mode, device)
         print(f"{mode}: train {results[mode]['loss_train'][0]:.5f} -> "
