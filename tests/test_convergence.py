@@ -13,26 +13,31 @@ and actually learn. Regenerate with
 import json
 import os
 
+import pytest
+
 HERE = os.path.dirname(os.path.abspath(__file__))
-PATH = os.path.join(HERE, "..", "profiles", "convergence_nbody.json")
 
 
-def _load():
-    with open(PATH) as f:
+def _load(workload):
+    with open(os.path.join(HERE, "..", "profiles",
+                           f"convergence_{workload}.json")) as f:
         return json.load(f)
 
 
-def test_convergence_curves_overlay():
-    d = _load()
+@pytest.mark.parametrize("workload,n_min,tol", [("nbody", 20, 0.05),
+                                                ("water3d", 15, 0.10)])
+def test_convergence_curves_overlay(workload, n_min, tol):
+    d = _load(workload)
     ref, fast = d["reference"]["loss_train"], d["fast"]["loss_train"]
-    assert len(ref) == len(fast) >= 20          # real epochs, not a smoke
+    assert len(ref) == len(fast) >= n_min       # real epochs, not a smoke
     assert d["device"].startswith("cuda")        # measured on the GPU
     for a, b in zip(fast, ref):
-        assert abs(a - b) / max(abs(b), 1e-9) < 0.05
+        assert abs(a - b) / max(abs(b), 1e-9) < tol
 
 
-def test_convergence_actually_learns():
-    d = _load()
+@pytest.mark.parametrize("workload", ["nbody", "water3d"])
+def test_convergence_actually_learns(workload):
+    d = _load(workload)
     for mode in ("reference", "fast"):
         tr = d[mode]["loss_train"]
         assert tr[-1] < 0.15 * tr[0]            # >85% train-loss reduction
