@@ -16,6 +16,7 @@ import torch.nn.functional as F
 from torch import nn
 
 from .. import ops
+from .. import ops as ops_mod
 
 
 def aggregate(message, row_index, n_node, aggr="sum", mask=None):
@@ -438,9 +439,9 @@ class PoolingNet(nn.Module):
 
 def _adj_matmul(edge_index, n, dense):
     """(unweighted adjacency) @ dense — replaces torch_sparse.spmm
-    (reference basic.py:663,668): row i accumulates dense[col[e]]."""
-    msg = dense.index_select(0, edge_index[1])
-    return aggregate(msg, edge_index[0], n, "sum")
+    (reference basic.py:663,668) with the deterministic CSR SpMM
+    composition (ops.spmm_adj: sort -> rowptr -> gather -> segment sum)."""
+    return ops_mod.spmm_adj(edge_index, n, dense)
 
 
 class EGHN(nn.Module):
