@@ -123,22 +123,25 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
     biases[2 * H + i] = b3[i];
     biases[3 * H + i] = w3v[i];
   }
+  __syncthreads();  // biases initialized by wave 0
 
+  // Wave-per-subtile execution (round 2, matching fused_edge_bwd): every
+  // phase reads/writes only its own wave's 16 rows, so the tile loop
+  // carries NO barriers — waves run independently.
   for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
-    __syncthreads();
     int* rws = reinterpret_cast<int*>(smem + L.rows);
     int* cls = reinterpret_cast<int*>(smem + L.cols);
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       rws[e] = e < nedge ? (int)row[e0 + e] : 0;
       cls[e] = e < nedge ? (int)col[e0 + e] : 0;
     }
-    __syncthreads();
 
-    // gather stage: in_tile [64][K_STRIDE]
-    for (int idx = tid; idx < TILE * 16; idx += THREADS) {
-      int e = idx / 16, piece = idx % 16;
+    // gather stage: in_tile [64][K_STRIDE] (wave-local rows)
+    for (int idx = lane; idx < 16 * 16; idx += 64) {
+      int e = wave * 16 + idx / 16, piece = idx % 16;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       int c8 = (piece & 7) * 8;
       bf16x8 v = {};
@@ -148,7 +151,8 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
     }
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float dx = 0, dy = 0, dz = 0, r2 = 0, a0 = 0, a1 = 0;
@@ -174,7 +178,6 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
 #pragma unroll
       for (int k = K_IN; k < K_PAD; ++k) brow[k] = (__bf16)0.f;
     }
-    __syncthreads();
 
     {  // GEMM1 -> t1
       f32x4 acc[4] = {};
@@ -191,7 +194,6 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
         }
       }
     }
-    __syncthreads();
     {  // GEMM2 -> msg
       f32x4 acc[4] = {};
       mm_a_lds<2>(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
@@ -207,7 +209,6 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
         }
       }
     }
-    __syncthreads();
     {  // GEMM3 + head -> p
       f32x4 acc[4] = {};
       mm_a_lds<2>(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
@@ -232,16 +233,16 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
           pv[wave * 16 + (lane >> 4) * 4 + r] = part[r];
       }
     }
-    __syncthreads();
 
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
+    for (int idx = lane; idx < 16 * 8; idx += 64) {
+      int e = wave * 16 + idx / 8;
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
           lds8(smem, L.msg + (e * H_STRIDE + c8) * 2);
     }
-    for (int e = tid; e < nedge; e += THREADS) {
+    if (lane < 16 && wave * 16 + lane < nedge) {
+      int e = wave * 16 + lane;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float p = reinterpret_cast<float*>(smem + L.pvec)[e];
       trans_out[(e0 + e) * 3] = dptr[0] * p;

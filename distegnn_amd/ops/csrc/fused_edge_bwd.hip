@@ -224,22 +224,31 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     gbacc[H + i] = 0.f;
     gbacc[2 * H + i] = 0.f;
   }
+  __syncthreads();  // biases/accumulators initialized by wave 0
 
+  // Wave-per-subtile execution (round 2): every phase below reads and
+  // writes ONLY its own wave's 16 rows of each LDS tile, so the plain
+  // variant needs NO intra-tile barriers at all — 12 independent waves
+  // per CU instead of 3 barrier-convoyed 4-wave groups (PMC showed 62%
+  // SQ_WAIT_ANY dominated by the 14 sync-separated phases). The FUSE_WG
+  // variant's wgrad contraction runs over the whole 64-edge tile, so it
+  // keeps the original barrier schedule.
   for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     int* rws = reinterpret_cast<int*>(smem + L.rows);
     int* cls = reinterpret_cast<int*>(smem + L.cols);
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       rws[e] = e < nedge ? (int)row[e0 + e] : 0;
       cls[e] = e < nedge ? (int)col[e0 + e] : 0;
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
 
-    // ---- stage ein ----
-    for (int idx = tid; idx < TILE * 16; idx += THREADS) {
-      int e = idx / 16, piece = idx % 16;
+    // ---- stage ein (wave-local rows) ----
+    for (int idx = lane; idx < 16 * 16; idx += 64) {
+      int e = wave * 16 + idx / 16, piece = idx % 16;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       int c8 = (piece & 7) * 8;
       bf16x8 v = {};
@@ -249,7 +258,8 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
     }
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float dx = 0, dy = 0, dz = 0, r2 = 0, a0 = 0, a1 = 0;
@@ -271,11 +281,11 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
 #pragma unroll
       for (int k = K_IN; k < K_PAD; ++k) brow[k] = (__bf16)0.f;
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     // ein -> global (only the split-K wgrad path consumes it)
     if constexpr (!FUSE_WG) {
-      for (int idx = tid; idx < TILE * (K_OUT / 8); idx += THREADS) {
-        int e = idx / (K_OUT / 8);
+      for (int idx = lane; idx < 16 * (K_OUT / 8); idx += 64) {
+        int e = wave * 16 + idx / (K_OUT / 8);
         if (e >= nedge) continue;
         int c8 = (idx % (K_OUT / 8)) * 8;
         *reinterpret_cast<bf16x8*>(ein_out + (e0 + e) * K_OUT + c8) =
@@ -299,11 +309,11 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
         }
       }
     }
-    __syncthreads();
-    // t1 = silu(z1) -> global (coalesced)
+    if constexpr (FUSE_WG) __syncthreads();
+    // t1 = silu(z1) -> global (coalesced, wave-local rows)
     if constexpr (!FUSE_WG) {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         if (e >= nedge) continue;
         int c8 = (idx % 8) * 8;
         *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
@@ -325,10 +335,10 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
         }
       }
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     if constexpr (!FUSE_WG) {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         if (e >= nedge) continue;
         int c8 = (idx % 8) * 8;
         *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
@@ -364,10 +374,11 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
           sc[(wave * 16 + (lane >> 4) * 4 + r) * 4] = part[r];  // p
       }
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
 
     // ---- head backward: dp; dw3v partial; dz3 overwrites z3 ----
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       float dp = 0.f;
@@ -379,7 +390,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       }
       sc[1] = dp;
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     {
       // thread covers column c over 16 edges: dw3v partial + dz3 in place
       int c = tid & 63;
@@ -399,7 +410,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       atomicAdd(&wpart[c], acc_w);
       atomicAdd(&gbacc[2 * H + c], acc_b3);
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     if constexpr (FUSE_WG) {
       // dW3 += dz3^T (in z3) @ silu(z2): z3 is consumed below (dmsg
       // staging overwrites it only after a barrier every wave reaches
@@ -407,8 +418,8 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       wg_acc<4, true>(smem, L.z3, H_STRIDE * 2, L.z2, H_STRIDE * 2, wave,
                       lane, wg3);
     } else {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         if (e >= nedge) continue;
         int c8 = (idx % 8) * 8;
         *reinterpret_cast<bf16x8*>(dz3_out + (e0 + e) * H + c8) =
@@ -421,17 +432,17 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       f32x4 acc[4] = {};
       mm_g<2, false>(smem, L.z3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w3t), H, lane, acc);
-      __syncthreads();
+      if constexpr (FUSE_WG) __syncthreads();
       // z3 consumed: reuse its tile to stage dmsg_n[row] COALESCED
       // (the C-layout merge otherwise issues 16 scattered 2 B loads/lane)
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         int c8 = (idx % 8) * 8;
         bf16x8 v = {};
         if (e < nedge) v = g8(dmsg_n + (long)rws[e] * H + c8);
         *reinterpret_cast<bf16x8*>(smem + L.z3 + (e * H_STRIDE + c8) * 2) = v;
       }
-      __syncthreads();
+      if constexpr (FUSE_WG) __syncthreads();
       __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
       const __bf16* dmsg_t = reinterpret_cast<const __bf16*>(smem + L.z3);
 #pragma unroll
@@ -446,15 +457,15 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
         }
       }
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     if constexpr (FUSE_WG) {
       // dW2 += dz2^T (in z2) @ silu(z1): z1 still holds pre-activations
       // (overwritten only after the barrier inside the dz1 phase)
       wg_acc<4, true>(smem, L.z2, H_STRIDE * 2, L.z1, H_STRIDE * 2, wave,
                       lane, wg2);
     } else {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         if (e >= nedge) continue;
         int c8 = (idx % 8) * 8;
         *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
@@ -476,7 +487,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       f32x4 acc[4] = {};
       mm_g<2, false>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w2t), H, lane, acc);
-      __syncthreads();
+      if constexpr (FUSE_WG) __syncthreads();
       __bf16* z1 = reinterpret_cast<__bf16*>(smem + L.z1);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -489,10 +500,10 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
         }
       }
     }
-    __syncthreads();
+    if constexpr (FUSE_WG) __syncthreads();
     if constexpr (!FUSE_WG) {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         if (e >= nedge) continue;
         int c8 = (idx % 8) * 8;
         *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
@@ -551,10 +562,10 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
         }
       }
     }
-    __syncthreads();
-    // dh_row / dh_col -> global (coalesced)
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
+    if constexpr (FUSE_WG) __syncthreads();
+    // dh_row / dh_col -> global (coalesced, wave-local rows)
+    for (int idx = lane; idx < 16 * 8; idx += 64) {
+      int e = wave * 16 + idx / 8;
       if (e >= nedge) continue;
       int c8 = (idx % 8) * 8;
       *reinterpret_cast<bf16x8*>(dhr_out + (e0 + e) * H + c8) =
@@ -562,8 +573,9 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       *reinterpret_cast<bf16x8*>(dhc_out + (e0 + e) * H + c8) =
           lds8(smem, L.in_tile + (e * K_STRIDE + H + c8) * 2);
     }
-    // dcd
-    for (int e = tid; e < nedge; e += THREADS) {
+    // dcd (wave-local rows)
+    if (lane < 16 && wave * 16 + lane < nedge) {
+      int e = wave * 16 + lane;
       long ge = e0 + e;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       const float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;

@@ -159,10 +159,12 @@ __host__ __device__ constexpr VSmemB vsmem_layout_bwd() {
     }                                                                       \
   }
 
-// copy an LDS H-tile to global [R,64] bf16 (optionally through silu)
+// copy an LDS H-tile to global [R,64] bf16 (optionally through silu).
+// Wave-local rows (wave-per-subtile execution, round 2): each wave copies
+// only its own 16 rows, so no barrier is needed around the copy.
 #define TILE_TO_GLOBAL(off, dst, SILU)                                      \
-  for (int idx = tid; idx < TILE * 8; idx += THREADS) {                     \
-    int e = idx / 8;                                                        \
+  for (int idx = lane; idx < 16 * 8; idx += 64) {                           \
+    int e = wave * 16 + idx / 8;                                            \
     if (e >= nrow) continue;                                                \
     int c8 = (idx % 8) * 8;                                                 \
     *reinterpret_cast<bf16x8*>((dst) + (r0 + e) * H + c8) =                 \
@@ -207,16 +209,18 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
     biases[4 * H + i] = wxvv[i];
     biases[5 * H + i] = wXv[i];
   }
+  __syncthreads();  // biases initialized by wave 0
 
+  // Wave-per-subtile execution (round 2): every phase touches only its
+  // own wave's 16 rows -> no intra-tile barriers.
   for (long tile = blockIdx.x; tile * TILE < n_rows; tile += gridDim.x) {
     long r0 = tile * TILE;
     int nrow = (int)((n_rows - r0 < (long)TILE) ? (n_rows - r0)
                                                 : (long)TILE);
-    __syncthreads();
 
-    // ---- gather vin ----
-    for (int idx = tid; idx < TILE * 16; idx += THREADS) {
-      int e = idx / 16, piece = idx % 16;
+    // ---- gather vin (wave-local rows) ----
+    for (int idx = lane; idx < 16 * 16; idx += 64) {
+      int e = wave * 16 + idx / 16, piece = idx % 16;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       int c8 = (piece & 7) * 8;
       bf16x8 v = {};
@@ -232,7 +236,8 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
       }
       *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
     }
-    for (int e = tid; e < TILE; e += THREADS) {
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
       float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       __bf16* brow = reinterpret_cast<__bf16*>(dst);
@@ -259,10 +264,9 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
         brow[2 * H + 1 + CMAX + k] = (__bf16)0.f;
       for (int j = cdim; j < CMAX; ++j) brow[2 * H + 1 + j] = (__bf16)0.f;
     }
-    __syncthreads();
     if (TRAIN) {
-      for (int idx = tid; idx < TILE * (K_OUT / 8); idx += THREADS) {
-        int e = idx / (K_OUT / 8);
+      for (int idx = lane; idx < 16 * (K_OUT / 8); idx += 64) {
+        int e = wave * 16 + idx / (K_OUT / 8);
         if (e >= nrow) continue;
         int c8 = (idx % (K_OUT / 8)) * 8;
         *reinterpret_cast<bf16x8*>(vin_out + (r0 + e) * K_OUT + c8) =
@@ -277,7 +281,6 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
       __bf16* za = reinterpret_cast<__bf16*>(smem + L.za);
       EPI_WRITE(za, x + biases[cc]);
     }
-    __syncthreads();
     if (TRAIN) TILE_TO_GLOBAL(L.za, z1_out, false);
     {  // z2
       f32x4 acc[4] = {};
@@ -286,7 +289,6 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
       __bf16* zb = reinterpret_cast<__bf16*>(smem + L.zb);
       EPI_WRITE(zb, x + biases[H + cc]);
     }
-    __syncthreads();
     if (TRAIN) TILE_TO_GLOBAL(L.zb, z2_out, false);
     TILE_TO_GLOBAL(L.zb, vmsg_out, true);
 
@@ -322,7 +324,6 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
         for (int rr = 0; rr < 4; ++rr)
           sc[(wave * 16 + (lane >> 4) * 4 + rr) * 4 + head] = part[rr];
       }
-      __syncthreads();
       if (TRAIN) {
         if (head == 0) {
           TILE_TO_GLOBAL(L.zc, zxv_out, false);
@@ -330,12 +331,11 @@ __global__ __launch_bounds__(THREADS) void fused_virtual_fwd(
           TILE_TO_GLOBAL(L.zc, zX_out, false);
         }
       }
-      // next head overwrites zc: wait for the copy readers
-      __syncthreads();
     }
 
-    // tv / tx / p2
-    for (int e = tid; e < nrow; e += THREADS) {
+    // tv / tx / p2 (wave-local rows)
+    if (lane < 16 && wave * 16 + lane < nrow) {
+      int e = wave * 16 + lane;
       const float* dptr = reinterpret_cast<float*>(smem + L.diff) + e * 4;
       const float* sc = reinterpret_cast<float*>(smem + L.scal) + e * 4;
       long r = r0 + e;
@@ -389,15 +389,18 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
     biases[H + i] = wXv[i];
     for (int kacc = 0; kacc < 6; ++kacc) gbacc[kacc * H + i] = 0.f;
   }
+  __syncthreads();  // biases/accumulators initialized by wave 0
 
+  // Wave-per-subtile execution (round 2): no intra-tile barriers — every
+  // phase touches only its own wave's 16 rows of each region.
   for (long tile = blockIdx.x; tile * TILE < n_rows; tile += gridDim.x) {
     long r0 = tile * TILE;
     int nrow = (int)((n_rows - r0 < (long)TILE) ? (n_rows - r0)
                                                 : (long)TILE);
-    __syncthreads();
 
-    // per-row: vdiff recompute, dpxv/dpX, initial dvd
-    for (int e = tid; e < TILE; e += THREADS) {
+    // per-row: vdiff recompute, dpxv/dpX, initial dvd (wave-local)
+    if (lane < 16) {
+      int e = wave * 16 + lane;
       float* dptr = reinterpret_cast<float*>(smem + LB.diff) + e * 4;
       float* sc = reinterpret_cast<float*>(smem + LB.scal) + e * 4;
       float dpxv = 0, dpX = 0;
@@ -420,14 +423,12 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
       sc[0] = dpxv;
       sc[1] = dpX;
     }
-    __syncthreads();
 
     // dzxv = dpxv * wxv o silu'(zxv); dzX likewise.
     // zxv/zX are staged COALESCED through the (currently free) za and
-    // in_tile regions — the C-layout loop otherwise issues 16 scattered
-    // 2 B global loads per lane per tensor.
-    for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-      int e = idx / 8;
+    // sa regions (wave-local rows).
+    for (int idx = lane; idx < 16 * 8; idx += 64) {
+      int e = wave * 16 + idx / 8;
       int c8 = (idx % 8) * 8;
       bf16x8 v = {}, w = {};
       if (e < nrow) {
@@ -438,7 +439,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
       *reinterpret_cast<bf16x8*>(smem + LB.sa
                                  + (e * H_STRIDE + c8) * 2) = w;
     }
-    __syncthreads();
     {
       int cc = tid & 63;
       int estart = (tid >> 6) * 16;
@@ -471,7 +471,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
       atomicAdd(&gbacc[4 * H + cc], awxvv);
       atomicAdd(&gbacc[5 * H + cc], awXv);
     }
-    __syncthreads();
     TILE_TO_GLOBAL(LB.zc, dzxv_out, false);
     TILE_TO_GLOBAL(LB.sa, dzX_out, false);
 
@@ -482,10 +481,9 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
                      opaque(wxvt), H, lane, acc);
       mm_g<2, false>(smem, LB.sa + wave * 16 * H_STRIDE * 2,
                      H_STRIDE * 2, opaque(wXt), H, lane, acc);
-      __syncthreads();
-      // in_tile consumed: re-stage za/in_tile with dvmsg and z2, coalesced
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      // sa consumed: re-stage za/sa with dvmsg and z2 (wave-local rows)
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         int c8 = (idx % 8) * 8;
         bf16x8 v = {}, w = {};
         if (e < nrow) {
@@ -496,7 +494,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
         *reinterpret_cast<bf16x8*>(smem + LB.sa
                                    + (e * H_STRIDE + c8) * 2) = w;
       }
-      __syncthreads();
       __bf16* zb = reinterpret_cast<__bf16*>(smem + LB.zb);
       const __bf16* ups = reinterpret_cast<const __bf16*>(smem + LB.za);
       const __bf16* z2s = reinterpret_cast<const __bf16*>(smem + LB.sa);
@@ -513,7 +510,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
         }
       }
     }
-    __syncthreads();
     TILE_TO_GLOBAL(LB.zb, dz2_out, false);
     {
       int cc = tid & 63;
@@ -528,8 +524,8 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
     // dz1 = (dz2 @ W2) o silu'(z1); z1 staged into za coalesced, each
     // slot read once then overwritten in place by its dz1 value
     {
-      for (int idx = tid; idx < TILE * 8; idx += THREADS) {
-        int e = idx / 8;
+      for (int idx = lane; idx < 16 * 8; idx += 64) {
+        int e = wave * 16 + idx / 8;
         int c8 = (idx % 8) * 8;
         bf16x8 v = {};
         if (e < nrow) v = g8(z1_in + (r0 + e) * H + c8);
@@ -538,7 +534,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
       f32x4 acc[4] = {};
       mm_g<2, false>(smem, LB.zb + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w2t), H, lane, acc);
-      __syncthreads();
       __bf16* za = reinterpret_cast<__bf16*>(smem + LB.za);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -551,7 +546,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
         }
       }
     }
-    __syncthreads();
     TILE_TO_GLOBAL(LB.za, dz1_out, false);
     {
       int cc = tid & 63;
@@ -585,7 +579,6 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
         }
       }
-      __syncthreads();  // prior pass's sa readers have drained
       __bf16* dvin = reinterpret_cast<__bf16*>(smem + LB.sa);
 #pragma unroll
       for (int nt = 0; nt < 3; ++nt) {
@@ -600,11 +593,11 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
           }
         }
       }
-      __syncthreads();
       // this pass's dh/dvf columns -> global (48-col windows stay
-      // 8-aligned; cols >= 2H are the vrad/dgram tail handled below)
-      for (int idx = tid; idx < TILE * 6; idx += THREADS) {
-        int e = idx / 6;
+      // 8-aligned; cols >= 2H are the vrad/dgram tail handled below;
+      // wave-local rows, so the passes need no barriers)
+      for (int idx = lane; idx < 16 * 6; idx += 64) {
+        int e = wave * 16 + idx / 6;
         if (e >= nrow) continue;
         int c8 = pass * 48 + (idx % 6) * 8;
         if (c8 >= 2 * H) continue;
@@ -618,7 +611,8 @@ __global__ __launch_bounds__(THREADS, 4) void fused_virtual_bwd(
 
     // dgram (logical cols 2H+1.. live in the pass-2 window at physical
     // offset -96) + dvd (per-row)
-    for (int e = tid; e < nrow; e += THREADS) {
+    if (lane < 16 && wave * 16 + lane < nrow) {
+      int e = wave * 16 + lane;
       long r = r0 + e;
       const __bf16* dvin = reinterpret_cast<const __bf16*>(
           smem + LB.sa) + e * H_STRIDE;
