@@ -49,37 +49,76 @@ __global__ __launch_bounds__(THREADS) void wgrad_splitk(
 
   f32x4 acc[4][NTW] = {};
 
-  for (long e0 = c0; e0 < c1; e0 += E_STEP) {
+  // Register double-buffering (round 2): the next round's global loads of
+  // g/x are issued DURING the current round's MFMA phase (they have no LDS
+  // dependency), hiding the global latency the old stage->barrier->mfma
+  // sequence exposed every E_STEP rows. Per-thread prefetch registers are
+  // sized by the template's column capacity (ip <= NTW*64).
+  constexpr int GITER = (E_STEP * (O_DIM / 8) + THREADS - 1) / THREADS;
+  constexpr int XITER = (E_STEP * (NTW * 64 / 8) + THREADS - 1) / THREADS;
+  bf16x8 pg[GITER];
+  bf16x8 px[XITER];
+
+  auto prefetch = [&](long e0) {
     int ne = (int)((c1 - e0 < E_STEP) ? (c1 - e0) : (long)E_STEP);
-    __syncthreads();
-    // stage g[e0..e0+31][0..63] transposed -> gT[o][e]
-    for (int idx = tid; idx < E_STEP * (O_DIM / 8); idx += THREADS) {
+#pragma unroll
+    for (int it = 0; it < GITER; ++it) {
+      int idx = tid + it * THREADS;
       int e = idx / (O_DIM / 8);
       int o8 = (idx % (O_DIM / 8)) * 8;
       bf16x8 v = {};
-      if (e < ne) v = *reinterpret_cast<const bf16x8*>(g + (e0 + e) * O_DIM + o8);
+      if (idx < E_STEP * (O_DIM / 8) && e < ne)
+        v = *reinterpret_cast<const bf16x8*>(g + (e0 + e) * O_DIM + o8);
+      pg[it] = v;
+    }
+#pragma unroll
+    for (int it = 0; it < XITER; ++it) {
+      int idx = tid + it * THREADS;
+      int e = idx / (ip / 8);
+      int i8 = (idx % (ip / 8)) * 8;
+      bf16x8 v = {};
+      if (idx < E_STEP * (ip / 8) && e < ne) {
+        if (i8 + 8 <= i_dim) {
+          v = *reinterpret_cast<const bf16x8*>(x + (e0 + e) * i_dim + i8);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) {
+            int i = i8 + u;
+            v[u] = i < i_dim ? ((const __bf16*)x)[(e0 + e) * i_dim + i]
+                             : (__bf16)0.f;
+          }
+        }
+      }
+      px[it] = v;
+    }
+  };
+
+  prefetch(c0);
+  for (long e0 = c0; e0 < c1; e0 += E_STEP) {
+    __syncthreads();
+    // LDS-transpose the prefetched round: gT[o][e], xT[i][e]
+#pragma unroll
+    for (int it = 0; it < GITER; ++it) {
+      int idx = tid + it * THREADS;
+      if (idx >= E_STEP * (O_DIM / 8)) break;
+      int e = idx / (O_DIM / 8);
+      int o8 = (idx % (O_DIM / 8)) * 8;
+      bf16x8 v = pg[it];
 #pragma unroll
       for (int u = 0; u < 8; ++u) gT[(o8 + u) * T_STRIDE + e] = v[u];
     }
-    // stage x[e][i] transposed -> xT[i][e] (i padded with zeros)
-    for (int idx = tid; idx < E_STEP * (ip / 8); idx += THREADS) {
+#pragma unroll
+    for (int it = 0; it < XITER; ++it) {
+      int idx = tid + it * THREADS;
+      if (idx >= E_STEP * (ip / 8)) break;
       int e = idx / (ip / 8);
       int i8 = (idx % (ip / 8)) * 8;
-      if (e < ne && i8 + 8 <= i_dim) {
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(x + (e0 + e) * i_dim + i8);
+      bf16x8 v = px[it];
 #pragma unroll
-        for (int u = 0; u < 8; ++u) xT[(i8 + u) * T_STRIDE + e] = v[u];
-      } else {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          int i = i8 + u;
-          __bf16 val = (__bf16)0.f;
-          if (e < ne && i < i_dim) val = ((const __bf16*)x)[(e0 + e) * i_dim + i];
-          xT[i * T_STRIDE + e] = val;
-        }
-      }
+      for (int u = 0; u < 8; ++u) xT[(i8 + u) * T_STRIDE + e] = v[u];
     }
     __syncthreads();
+    if (e0 + E_STEP < c1) prefetch(e0 + E_STEP);
     // D[o][i] += gT[o][e] * xT[i][e] — A rows = o, B cols = i, K = e
 #pragma unroll
     for (int kk = 0; kk < E_STEP / 32; ++kk) {
