@@ -30,7 +30,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // NTW = n-tiles (of 16 cols of x) per wave; ITILES total = waves used * NTW
 // mapping: wave w covers n-tiles [w*NTW, w*NTW+NTW)
 template <int NTW, int E_STEP, int T_STRIDE>
-__global__ __launch_bounds__(THREADS) void wgrad_splitk(
+__global__ __launch_bounds__(THREADS, NTW <= 2 ? 4 : 3) void wgrad_splitk(
     const bf16* __restrict__ g,  // [M, 64]
     const bf16* __restrict__ x,  // [M, I]
     float* __restrict__ part,    // [nchunk, 64, IP] (IP = 16*ceil(I/16))
