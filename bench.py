@@ -290,7 +290,15 @@ def main():
                                    rank=rank)
 
     mse = None
-    for w in range(args.warmup):
+    # a shape's graph is captured at its (warmup_occurrences+1)-th
+    # occurrence; a capture takes SECONDS at large node counts, so the
+    # warmup phase must cover capture + at least one replay for EVERY
+    # batch shape no matter how small --warmup is (extra steps stay
+    # untimed — only the K steps below are measured)
+    n_warm = args.warmup
+    if graphed.enabled:
+        n_warm = max(n_warm, (graphed.warmup + 2) * len(batches))
+    for w in range(n_warm):
         mse = train_step(graphed, batches[w % len(batches)], optimizer,
                          grad_bucket, w, accum, ws_eff, device, clip,
                          mmd_cfg=num_sample)
@@ -315,7 +323,7 @@ def main():
     t0 = time.perf_counter()
     for k in range(args.steps):
         mse = train_step(graphed, batches[k % len(batches)], optimizer,
-                         grad_bucket, args.warmup + k, accum,
+                         grad_bucket, n_warm + k, accum,
                          ws_eff, device, clip, mmd_cfg=num_sample)
     if event_timing:
         ev1.record()
