@@ -27,3 +27,15 @@ class EdgeGraph:
     @property
     def num_edges(self):
         return self.edge_index.size(1)
+
+    def dst_csr(self):
+        """(dstptr [N+1], perm [M]) with dst[perm] sorted — cached; feeds
+        the fused CSR edge_softmax kernel."""
+        if getattr(self, "_dst_csr", None) is None:
+            perm = torch.argsort(self.dst, stable=True)
+            ptr = torch.zeros(self.num_nodes + 1, dtype=torch.long,
+                              device=self.dst.device)
+            ptr.scatter_add_(0, self.dst.index_select(0, perm) + 1,
+                             torch.ones_like(perm))
+            self._dst_csr = (ptr.cumsum(0), perm)
+        return self._dst_csr

@@ -35,9 +35,46 @@ def _segment_sum(data, index, n):
     return out.reshape(n, *shape[1:])
 
 
+class _EdgeSoftmaxFn(torch.autograd.Function):
+    """Fused CSR edge softmax (csrc/segment_reduce.hip edge_softmax_csr):
+    one wave per destination segment, max/sum-exp/write passes in
+    registers. Backward uses the standard softmax identity with a CSR
+    segment sum."""
+
+    @staticmethod
+    def forward(ctx, scores, dst, dstptr, perm, num_nodes):
+        from ... import ops as _ops
+
+        attn = _ops.hip_ext().edge_softmax_fwd(
+            scores.float().contiguous(), dstptr, perm)
+        ctx.save_for_backward(attn, dst)
+        ctx.n = num_nodes
+        return attn.to(scores.dtype)
+
+    @staticmethod
+    def backward(ctx, g):
+        attn, dst = ctx.saved_tensors
+        ag = attn * g.float()
+        s = _segment_sum(ag, dst, ctx.n)
+        dscores = ag - attn * s.index_select(0, dst)
+        return dscores.to(g.dtype), None, None, None, None
+
+
 def edge_softmax(scores: torch.Tensor, dst: torch.Tensor,
-                 num_nodes: int) -> torch.Tensor:
-    """Softmax over each node's incoming edges (DGL edge_softmax)."""
+                 num_nodes: int, csr=None) -> torch.Tensor:
+    """Softmax over each node's incoming edges (DGL edge_softmax).
+
+    With ``csr=(dstptr, perm)`` on GPU (power-of-two head count), the
+    fused CSR kernel runs; otherwise the scatter/exp/segment composition.
+    """
+    from ... import ops as _ops
+
+    h = scores.shape[-1]
+    if (csr is not None and scores.is_cuda and scores.dim() == 2
+            and 1 <= h <= 64 and (h & (h - 1)) == 0
+            and _ops.hip_ext() is not None
+            and hasattr(_ops.hip_ext(), "edge_softmax_fwd")):
+        return _EdgeSoftmaxFn.apply(scores, dst, csr[0], csr[1], num_nodes)
     # segment max for stability
     mx = torch.full((num_nodes,) + scores.shape[1:], -torch.inf,
                     dtype=scores.dtype, device=scores.device)
@@ -264,7 +301,8 @@ class GMABSE3(nn.Module):
         queries = fiber2head(q, h, self.f_key, squeeze=True)    # [N, h, f]
         scores = (keys * queries.index_select(0, G.dst)).sum(-1)  # [M, h]
         scores = scores / np.sqrt(self.f_key.n_features)
-        attn = edge_softmax(scores, G.dst, G.num_nodes)          # [M, h]
+        attn = edge_softmax(scores, G.dst, G.num_nodes,
+                            csr=G.dst_csr() if scores.is_cuda else None)
         out = {}
         for m, d in self.f_value.structure:
             weighted = attn.unsqueeze(-1).unsqueeze(-1) * values[d]

@@ -26,6 +26,8 @@ torch::Tensor coord_update_forward(torch::Tensor coord, torch::Tensor agg,
                                    torch::Tensor trans_v,
                                    torch::Tensor phiv, torch::Tensor vel);
 torch::Tensor coord_update_backward(torch::Tensor g, torch::Tensor vel);
+torch::Tensor edge_softmax_fwd(torch::Tensor scores, torch::Tensor dstptr,
+                               torch::Tensor perm);
 torch::Tensor cfconv_forward(torch::Tensor xw1, torch::Tensor dist,
                              torch::Tensor row, torch::Tensor col,
                              torch::Tensor w1f, torch::Tensor b1,
@@ -134,6 +136,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused coord + agg + trans_v + phi_v*vel");
   m.def("coord_update_backward", &coord_update_backward,
         "dphiv = sum_d g*vel for the fused coordinate update");
+  m.def("edge_softmax_fwd", &edge_softmax_fwd,
+        "CSR edge softmax over destination segments (SE(3) attention)",
+        py::arg("scores"), py::arg("dstptr"), py::arg("perm"));
   m.def("cfconv_forward", &cfconv_forward,
         "fused SchNet CFConv messages: smearing + filter MLP + cutoff + "
         "gathered multiply");
