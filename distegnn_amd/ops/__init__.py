@@ -531,7 +531,10 @@ def _warn_fallback(op: str, reason: str):
     print(f"[distegnn_amd.ops] {op}: fused MFMA kernel not applicable "
           f"({reason}); running the eager composition instead — expect a "
           f"large per-step slowdown. The hand-written gfx950 kernels cover "
-          f"hidden_nf=64, edge_attr_nf=2, virtual_channels<=8, bf16.",
+          f"hidden_nf=64, edge_attr_nf=2, virtual_channels<=8, bf16. "
+          f"(fp32 evaluation is the reference-parity default and is "
+          f"expected to take this path; set train.bf16_eval=true to "
+          f"evaluate on the fused bf16 kernels.)",
           flush=True)
 
 
