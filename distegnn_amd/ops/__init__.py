@@ -466,6 +466,7 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
             # 24.3 ms/step net). Off by default until the LDS-phase
             # pipeline hides enough latency at 2 waves to win.
             use_wg = (hasattr(ext, "fused_edge_backward_wg")
+                      and h.size(1) == 64
                       and os.environ.get("DISTEGNN_EDGE_WGRAD_FUSED",
                                          "0") == "1")
             if use_wg:
@@ -531,7 +532,8 @@ def _warn_fallback(op: str, reason: str):
     print(f"[distegnn_amd.ops] {op}: fused MFMA kernel not applicable "
           f"({reason}); running the eager composition instead — expect a "
           f"large per-step slowdown. The hand-written gfx950 kernels cover "
-          f"hidden_nf=64, edge_attr_nf=2, virtual_channels<=8, bf16. "
+          f"hidden_nf in {{32, 64, 128}} (edge; the virtual block needs "
+          f"64), edge_attr_nf=2, virtual_channels<=8, bf16. "
           f"(fp32 evaluation is the reference-parity default and is "
           f"expected to take this path; set train.bf16_eval=true to "
           f"evaluate on the fused bf16 kernels.)",
@@ -545,8 +547,9 @@ def _edge_fallback_reason(h, eattr, rowptr, colptr, col_perm):
         return None            # explicit opt-out
     if h.dtype != torch.bfloat16:
         return f"dtype {h.dtype} (kernels are bf16-in/fp32-accum)"
-    if h.size(1) != 64:
-        return f"hidden_nf={h.size(1)} (kernels are compiled for H=64)"
+    if h.size(1) not in (32, 64, 128):
+        return (f"hidden_nf={h.size(1)} (the edge kernels are compiled "
+                f"for H in {{32, 64, 128}})")
     if eattr is None or eattr.size(1) != 2:
         return (f"edge_attr_nf="
                 f"{0 if eattr is None else eattr.size(1)} (kernels expect 2)")
@@ -561,7 +564,8 @@ def fused_edge_block(h, coord, eattr, row, col, rowptr, colptr, col_perm,
 
     Returns (agg_msg [N,64], agg_trans [N,3]) — per-node MEANS of the edge
     messages and coordinate translations."""
-    usable = (h.is_cuda and h.dtype == torch.bfloat16 and h.size(1) == 64
+    usable = (h.is_cuda and h.dtype == torch.bfloat16
+              and h.size(1) in (32, 64, 128)
               and eattr is not None and eattr.size(1) == 2
               and rowptr is not None and colptr is not None
               and col_perm is not None and hip_ext() is not None

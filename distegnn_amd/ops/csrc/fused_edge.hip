@@ -1,7 +1,7 @@
 // Fused FastEGNN edge block — the north-star MFMA kernel (gfx950).
 //
 // Per edge e = (i=row[e], j=col[e]) this computes, in ONE kernel:
-//   in_e   = [h_i | h_j | r_e | a_e]            (K = 2H+1+Ea = 131, H=64)
+//   in_e   = [h_i | h_j | r_e | a_e]            (K = 2H+1+Ea)
 //   r_e    = ||x_i - x_j||^2, d_e = x_i - x_j   (optionally normalized)
 //   t1     = SiLU(in_e  @ W1^T + b1)            [H]
 //   msg    = SiLU(t1    @ W2^T + b2)            [H]   (edge_feat, phi_e)
@@ -11,12 +11,18 @@
 // writing msg [M,H] bf16 + trans [M,3] f32 (aggregated by the CSR
 // segment-mean kernels). All [M,.] intermediates stay in LDS/registers.
 //
-// Occupancy design (PMC-driven): the first version staged W1/W2/W3 in LDS
-// (81 KB) and ran at 1 block/CU with 65% of wave cycles parked in
-// SQ_WAIT_ANY. B-fragments are 16-byte k-contiguous reads, so the weights
-// (<=40 KB, L2-resident, shared by every block) are now read DIRECTLY from
-// global memory from row-padded copies prepared in the launcher; LDS holds
-// only the per-tile input/activation buffers (~43 KB -> 3 blocks/CU).
+// Templated over H in {32, 64, 128} (round 2): tile sizes, LDS strides
+// and MFMA n-tile counts derive from H; H=64 is the tuned headline shape
+// (the numbers in profiles/ are measured there), H=32/128 make any
+// hidden_nf the YAML surface allows run the MFMA path instead of
+// silently dropping to eager.
+//
+// Occupancy design (PMC-driven): weights (<=40 KB at H=64, L2-resident,
+// shared by every block) are read DIRECTLY from global memory from
+// row-padded copies prepared in the launcher; LDS holds only the
+// per-tile input/activation buffers. Wave-per-subtile execution: every
+// phase reads/writes only its own wave's 16 rows, so the tile loop
+// carries NO barriers.
 
 #include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
@@ -25,12 +31,7 @@
 
 namespace {
 
-constexpr int H = 64;
 constexpr int EA = 2;
-constexpr int K_IN = 2 * H + 1 + EA;  // 131
-constexpr int K_PAD = 160;            // 5 MFMA k-steps of 32
-constexpr int K_STRIDE = 168;         // LDS row stride (bank-free b128)
-constexpr int H_STRIDE = 72;
 constexpr int TILE = 64;
 constexpr int THREADS = 256;
 
@@ -42,6 +43,19 @@ __device__ __forceinline__ float silu(float x) {
   return x / (1.f + __expf(-x));
 }
 
+// per-H derived dimensions (K_PAD multiple of the 32-wide MFMA k-step;
+// +8-element strides keep the 16 b128 fragment readers on distinct banks)
+template <int H>
+struct ED {
+  static constexpr int K_IN = 2 * H + 1 + EA;
+  static constexpr int K_PAD = (K_IN + 31) / 32 * 32;
+  static constexpr int K_STRIDE = K_PAD + 8;
+  static constexpr int H_STRIDE = H + 8;
+  static constexpr int NT = H / 16;     // 16-col MFMA n-tiles per row
+  static constexpr int HP = H / 8;      // bf16x8 fragments per H row
+};
+
+template <int H>
 struct SmemLayout {
   int in_tile;  // [TILE][K_STRIDE] bf16
   int t1;       // [TILE][H_STRIDE] bf16
@@ -54,12 +68,13 @@ struct SmemLayout {
   int total;
 };
 
-__host__ __device__ constexpr SmemLayout smem_layout() {
-  SmemLayout L{};
+template <int H>
+__host__ __device__ constexpr SmemLayout<H> smem_layout() {
+  SmemLayout<H> L{};
   int o = 0;
-  L.in_tile = o; o += TILE * K_STRIDE * 2;
-  L.t1 = o; o += TILE * H_STRIDE * 2;
-  L.msg = o; o += TILE * H_STRIDE * 2;
+  L.in_tile = o; o += TILE * ED<H>::K_STRIDE * 2;
+  L.t1 = o; o += TILE * ED<H>::H_STRIDE * 2;
+  L.msg = o; o += TILE * ED<H>::H_STRIDE * 2;
   L.diff = o; o += TILE * 4 * 4;
   L.pvec = o; o += TILE * 4;
   L.bias = o; o += 4 * H * 4;
@@ -81,37 +96,44 @@ __device__ __forceinline__ const bf16* opaque(const bf16* p) {
   return p;
 }
 
-// 16(edges) x 64 GEMM: A from LDS, B (weights, k-contig rows [64][kb])
+// 16(edges) x H GEMM: A from LDS, B (weights, k-contig rows [H][kb])
 // from GLOBAL (L2-resident).
-template <int KSTEPS>
+template <int KSTEPS, int NT>
 __device__ __forceinline__ void mm_a_lds(const char* smem, int a_off,
                                          int a_stride,
                                          const bf16* __restrict__ w, int wk,
-                                         int lane, f32x4 (&acc)[4]) {
+                                         int lane, f32x4 (&acc)[NT]) {
 #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
     bf16x8 a = lds8(smem, a_off + (lane & 15) * a_stride + k * 2);
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
+    for (int nt = 0; nt < NT; ++nt) {
       bf16x8 b = g8(w + (nt * 16 + (lane & 15)) * wk + k);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
 }
 
-__global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
+template <int H>
+__global__ __launch_bounds__(THREADS, H <= 64 ? 2 : 1) void fused_edge_fwd(
     const bf16* __restrict__ h, const float* __restrict__ coord,
     const float* __restrict__ eattr, const long* __restrict__ row,
     const long* __restrict__ col,
-    const bf16* __restrict__ w1p,  // [64][K_PAD] padded
+    const bf16* __restrict__ w1p,  // [H][K_PAD] padded
     const float* __restrict__ b1, const bf16* __restrict__ w2,
     const float* __restrict__ b2, const bf16* __restrict__ w3,
     const float* __restrict__ b3, const float* __restrict__ w3v,
     bf16* __restrict__ msg_out, float* __restrict__ trans_out, long m,
     int normalize, float eps) {
+  constexpr int K_IN = ED<H>::K_IN;
+  constexpr int K_PAD = ED<H>::K_PAD;
+  constexpr int K_STRIDE = ED<H>::K_STRIDE;
+  constexpr int H_STRIDE = ED<H>::H_STRIDE;
+  constexpr int NT = ED<H>::NT;
+  constexpr int HP = ED<H>::HP;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr SmemLayout L = smem_layout();
+  constexpr SmemLayout<H> L = smem_layout<H>();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -125,9 +147,8 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
   }
   __syncthreads();  // biases initialized by wave 0
 
-  // Wave-per-subtile execution (round 2, matching fused_edge_bwd): every
-  // phase reads/writes only its own wave's 16 rows, so the tile loop
-  // carries NO barriers — waves run independently.
+  // Wave-per-subtile execution: every phase reads/writes only its own
+  // wave's 16 rows, so the tile loop carries NO barriers.
   for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
@@ -140,16 +161,16 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
     }
 
     // gather stage: in_tile [64][K_STRIDE] (wave-local rows)
-    for (int idx = lane; idx < 16 * 16; idx += 64) {
-      int e = wave * 16 + idx / 16, piece = idx % 16;
+    for (int idx = lane; idx < 16 * 2 * HP; idx += 64) {
+      int e = wave * 16 + idx / (2 * HP), piece = idx % (2 * HP);
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
-      int c8 = (piece & 7) * 8;
+      int c8 = (piece % HP) * 8;
       bf16x8 v = {};
       if (e < nedge) {
-        long src = piece < 8 ? rws[e] : cls[e];
+        long src = piece < HP ? rws[e] : cls[e];
         v = g8(h + src * H + c8);
       }
-      *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
+      *reinterpret_cast<bf16x8*>(dst + (piece < HP ? c8 : H + c8) * 2) = v;
     }
     if (lane < 16) {
       int e = wave * 16 + lane;
@@ -180,12 +201,12 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
     }
 
     {  // GEMM1 -> t1
-      f32x4 acc[4] = {};
-      mm_a_lds<K_PAD / 32>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
-                           K_STRIDE * 2, opaque(w1p), K_PAD, lane, acc);
+      f32x4 acc[NT] = {};
+      mm_a_lds<K_PAD / 32, NT>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
+                               K_STRIDE * 2, opaque(w1p), K_PAD, lane, acc);
       __bf16* t1 = reinterpret_cast<__bf16*>(smem + L.t1);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -195,12 +216,12 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       }
     }
     {  // GEMM2 -> msg
-      f32x4 acc[4] = {};
-      mm_a_lds<2>(smem, L.t1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                  opaque(w2), H, lane, acc);
+      f32x4 acc[NT] = {};
+      mm_a_lds<H / 32, NT>(smem, L.t1 + wave * 16 * H_STRIDE * 2,
+                           H_STRIDE * 2, opaque(w2), H, lane, acc);
       __bf16* mg = reinterpret_cast<__bf16*>(smem + L.msg);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -210,12 +231,12 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       }
     }
     {  // GEMM3 + head -> p
-      f32x4 acc[4] = {};
-      mm_a_lds<2>(smem, L.msg + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
-                  opaque(w3), H, lane, acc);
+      f32x4 acc[NT] = {};
+      mm_a_lds<H / 32, NT>(smem, L.msg + wave * 16 * H_STRIDE * 2,
+                           H_STRIDE * 2, opaque(w3), H, lane, acc);
       float part[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
         float wv = biases[3 * H + c];
         float bb = biases[2 * H + c];
@@ -234,10 +255,10 @@ __global__ __launch_bounds__(THREADS, 2) void fused_edge_fwd(
       }
     }
 
-    for (int idx = lane; idx < 16 * 8; idx += 64) {
-      int e = wave * 16 + idx / 8;
+    for (int idx = lane; idx < 16 * HP; idx += 64) {
+      int e = wave * 16 + idx / HP;
       if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
+      int c8 = (idx % HP) * 8;
       *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
           lds8(smem, L.msg + (e * H_STRIDE + c8) * 2);
     }
@@ -281,31 +302,34 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     std::vector<torch::Tensor> prepped) {
   TORCH_CHECK(h.is_cuda() && h.scalar_type() == torch::kBFloat16,
               "h must be CUDA bf16");
-  TORCH_CHECK(h.size(1) == H, "fused edge kernel requires hidden_nf=64");
+  long hdim = h.size(1);
+  TORCH_CHECK(hdim == 32 || hdim == 64 || hdim == 128,
+              "fused edge kernel supports hidden_nf in {32, 64, 128}");
   TORCH_CHECK(eattr.size(1) == EA, "fused edge kernel requires edge_attr_nf=2");
   auto hc = h.contiguous();
   auto cc = coord.contiguous().to(torch::kFloat);
   auto ec = eattr.contiguous().to(torch::kFloat);
   long m = row.numel();
-  auto msg = torch::empty({m, (long)H}, h.options());
+  auto msg = torch::empty({m, hdim}, h.options());
   auto trans = torch::empty({m, 3}, cc.options());
   if (m == 0) return {msg, trans};
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr SmemLayout L = smem_layout();
   long tiles = (m + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
   // Weight prep: python may pass version-cached transformed copies
   // (ops/prep.py) so replays/graphs skip ~7 tiny cast/pad kernels per
   // call; otherwise transform here.
   torch::Tensor w1p, w2c, w3c, b1c, b2c, b3c, w3vc;
+  long k_in = 2 * hdim + 1 + EA;
+  long k_pad = (k_in + 31) / 32 * 32;
   if (!prepped.empty()) {
     TORCH_CHECK(prepped.size() == 7, "edge fwd prepped wants 7 tensors");
     w1p = prepped[0]; w2c = prepped[1]; w3c = prepped[2];
     b1c = prepped[3]; b2c = prepped[4]; b3c = prepped[5]; w3vc = prepped[6];
-    TORCH_CHECK(w1p.size(1) == K_PAD, "prepped w1p must be row-padded");
+    TORCH_CHECK(w1p.size(1) == k_pad, "prepped w1p must be row-padded");
   } else {
-    // row-pad W1 to [64][K_PAD] so 16-B B-fragment reads are aligned
-    w1p = torch::constant_pad_nd(w1.contiguous(), {0, K_PAD - K_IN});
+    // row-pad W1 to [H][K_PAD] so 16-B B-fragment reads are aligned
+    w1p = torch::constant_pad_nd(w1.contiguous(), {0, k_pad - k_in});
     w2c = w2.contiguous();
     w3c = w3.contiguous();
     b1c = b1.contiguous().to(torch::kFloat);
@@ -313,16 +337,21 @@ std::tuple<torch::Tensor, torch::Tensor> fused_edge_forward(
     b3c = b3.contiguous().to(torch::kFloat);
     w3vc = w3v.contiguous().to(torch::kFloat);
   }
-  fused_edge_fwd<<<blocks, THREADS, L.total, stream>>>(
-      reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
-      ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
-      col.contiguous().data_ptr<long>(),
-      reinterpret_cast<const bf16*>(w1p.data_ptr()), b1c.data_ptr<float>(),
-      reinterpret_cast<const bf16*>(w2c.data_ptr()), b2c.data_ptr<float>(),
-      reinterpret_cast<const bf16*>(w3c.data_ptr()), b3c.data_ptr<float>(),
-      w3vc.data_ptr<float>(),
-      reinterpret_cast<bf16*>(msg.data_ptr()), trans.data_ptr<float>(), m,
-      normalize ? 1 : 0, (float)eps);
+#define LAUNCH_FWD(HH)                                                      \
+  fused_edge_fwd<HH><<<blocks, THREADS, smem_layout<HH>().total, stream>>>( \
+      reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),   \
+      ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),              \
+      col.contiguous().data_ptr<long>(),                                    \
+      reinterpret_cast<const bf16*>(w1p.data_ptr()), b1c.data_ptr<float>(), \
+      reinterpret_cast<const bf16*>(w2c.data_ptr()), b2c.data_ptr<float>(), \
+      reinterpret_cast<const bf16*>(w3c.data_ptr()), b3c.data_ptr<float>(), \
+      w3vc.data_ptr<float>(),                                               \
+      reinterpret_cast<bf16*>(msg.data_ptr()), trans.data_ptr<float>(), m,  \
+      normalize ? 1 : 0, (float)eps)
+  if (hdim == 32) LAUNCH_FWD(32);
+  else if (hdim == 64) LAUNCH_FWD(64);
+  else LAUNCH_FWD(128);
+#undef LAUNCH_FWD
   return {msg, trans};
 }
 

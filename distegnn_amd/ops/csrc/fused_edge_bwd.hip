@@ -27,15 +27,23 @@
 
 namespace {
 
-constexpr int H = 64;
 constexpr int EA = 2;
-constexpr int K_IN = 2 * H + 1 + EA;  // 131
-constexpr int K_PAD = 160;
-constexpr int K_STRIDE = 168;
-constexpr int K_OUT = 144;  // ein cols written to global (for dW1 GEMM)
-constexpr int H_STRIDE = 72;
 constexpr int TILE = 64;
 constexpr int THREADS = 256;
+
+// per-H derived dimensions (H=64 reproduces the tuned round-1 constants:
+// K_IN 131, K_PAD 160, K_STRIDE 168, K_OUT 144, H_STRIDE 72)
+template <int H>
+struct EDB {
+  static constexpr int K_IN = 2 * H + 1 + EA;
+  static constexpr int K_PAD = (K_IN + 31) / 32 * 32;
+  static constexpr int K_STRIDE = K_PAD + 8;
+  static constexpr int K_OUT = (K_IN + 15) / 16 * 16;  // ein cols (dW1 GEMM)
+  static constexpr int H_STRIDE = H + 8;
+  static constexpr int NT = H / 16;
+  static constexpr int HP = H / 8;
+  static constexpr int NTK = K_OUT / 16;               // dein n-tiles
+};
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
@@ -49,6 +57,7 @@ __device__ __forceinline__ float dsilu_(float x) {
   return s * (1.f + x * (1.f - s));
 }
 
+template <int H>
 struct Smem {
   int in_tile;  // [TILE][K_STRIDE] bf16  (ein, later dein)
   int z1;       // [TILE][H_STRIDE] bf16  (pre-act; overwritten by dz1)
@@ -64,13 +73,14 @@ struct Smem {
   int total;
 };
 
-__host__ __device__ constexpr Smem smem_layout() {
-  Smem L{};
+template <int H>
+__host__ __device__ constexpr Smem<H> smem_layout() {
+  Smem<H> L{};
   int o = 0;
-  L.in_tile = o; o += TILE * K_STRIDE * 2;
-  L.z1 = o; o += TILE * H_STRIDE * 2;
-  L.z2 = o; o += TILE * H_STRIDE * 2;
-  L.z3 = o; o += TILE * H_STRIDE * 2;
+  L.in_tile = o; o += TILE * EDB<H>::K_STRIDE * 2;
+  L.z1 = o; o += TILE * EDB<H>::H_STRIDE * 2;
+  L.z2 = o; o += TILE * EDB<H>::H_STRIDE * 2;
+  L.z3 = o; o += TILE * EDB<H>::H_STRIDE * 2;
   L.diff = o; o += TILE * 4 * 4;
   L.scal = o; o += TILE * 4 * 4;
   L.bias = o; o += 4 * H * 4;
@@ -153,18 +163,18 @@ __device__ __forceinline__ void wg_acc(const char* smem, int a_base,
 }
 
 // A from LDS (optionally through silu), B from global [64][wk] k-contig.
-template <int KSTEPS, bool SILU_A>
+template <int KSTEPS, bool SILU_A, int NT>
 __device__ __forceinline__ void mm_g(const char* smem, int a_off,
                                      int a_stride,
                                      const bf16* __restrict__ w, int wk,
-                                     int lane, f32x4 (&acc)[4]) {
+                                     int lane, f32x4 (&acc)[NT]) {
 #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
     bf16x8 a = SILU_A ? lds8_silu(smem, a_off + (lane & 15) * a_stride + k * 2)
                       : lds8(smem, a_off + (lane & 15) * a_stride + k * 2);
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
+    for (int nt = 0; nt < NT; ++nt) {
       bf16x8 b = g8(w + (nt * 16 + (lane & 15)) * wk + k);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
@@ -178,8 +188,10 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
 // the split-K wgrad kernels re-reading them). Costs 68 persistent
 // VGPRs -> 2 waves/SIMD instead of 3; saves the whole wgrad kernel
 // family plus the traffic.
-template <bool FUSE_WG>
-__global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
+template <int H, bool FUSE_WG>
+__global__ __launch_bounds__(THREADS,
+                             FUSE_WG ? 2 : (H <= 64 ? 3 : 1))
+void fused_edge_bwd(
     const bf16* __restrict__ h, const float* __restrict__ coord,
     const float* __restrict__ eattr, const long* __restrict__ row,
     const long* __restrict__ col,
@@ -201,8 +213,18 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     float* __restrict__ gw2_out,  // [H][H]
     float* __restrict__ gw3_out,  // [H][H]
     long m, int normalize, float eps) {
+  static_assert(!FUSE_WG || H == 64,
+                "the wgrad-fused variant is tuned for H=64 only");
+  constexpr int K_IN = EDB<H>::K_IN;
+  constexpr int K_PAD = EDB<H>::K_PAD;
+  constexpr int K_STRIDE = EDB<H>::K_STRIDE;
+  constexpr int K_OUT = EDB<H>::K_OUT;
+  constexpr int H_STRIDE = EDB<H>::H_STRIDE;
+  constexpr int NT = EDB<H>::NT;
+  constexpr int HP = EDB<H>::HP;
+  constexpr int NTK = EDB<H>::NTK;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr Smem L = smem_layout();
+  constexpr Smem<H> L = smem_layout<H>();
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -247,16 +269,16 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     if constexpr (FUSE_WG) __syncthreads();
 
     // ---- stage ein (wave-local rows) ----
-    for (int idx = lane; idx < 16 * 16; idx += 64) {
-      int e = wave * 16 + idx / 16, piece = idx % 16;
+    for (int idx = lane; idx < 16 * 2 * HP; idx += 64) {
+      int e = wave * 16 + idx / (2 * HP), piece = idx % (2 * HP);
       char* dst = smem + L.in_tile + e * K_STRIDE * 2;
-      int c8 = (piece & 7) * 8;
+      int c8 = (piece % HP) * 8;
       bf16x8 v = {};
       if (e < nedge) {
-        long src = piece < 8 ? rws[e] : cls[e];
+        long src = piece < HP ? rws[e] : cls[e];
         v = g8(h + src * H + c8);
       }
-      *reinterpret_cast<bf16x8*>(dst + (piece < 8 ? c8 : H + c8) * 2) = v;
+      *reinterpret_cast<bf16x8*>(dst + (piece < HP ? c8 : H + c8) * 2) = v;
     }
     if (lane < 16) {
       int e = wave * 16 + lane;
@@ -295,12 +317,12 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
 
     // ---- recompute: z1, z2, z3 (pre-activations) ----
     {
-      f32x4 acc[4] = {};
-      mm_g<K_PAD / 32, false>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
+      f32x4 acc[NT] = {};
+      mm_g<K_PAD / 32, false, NT>(smem, L.in_tile + wave * 16 * K_STRIDE * 2,
                               K_STRIDE * 2, opaque(w1p), K_PAD, lane, acc);
       __bf16* z1 = reinterpret_cast<__bf16*>(smem + L.z1);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -312,21 +334,21 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     if constexpr (FUSE_WG) __syncthreads();
     // t1 = silu(z1) -> global (coalesced, wave-local rows)
     if constexpr (!FUSE_WG) {
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
         if (e >= nedge) continue;
-        int c8 = (idx % 8) * 8;
+        int c8 = (idx % HP) * 8;
         *reinterpret_cast<bf16x8*>(t1_out + (e0 + e) * H + c8) =
             lds8_silu(smem, L.z1 + (e * H_STRIDE + c8) * 2);
       }
     }
     {
-      f32x4 acc[4] = {};
-      mm_g<2, true>(smem, L.z1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      f32x4 acc[NT] = {};
+      mm_g<H / 32, true, NT>(smem, L.z1 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                     opaque(w2), H, lane, acc);
       __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -337,22 +359,22 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     }
     if constexpr (FUSE_WG) __syncthreads();
     if constexpr (!FUSE_WG) {
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
         if (e >= nedge) continue;
-        int c8 = (idx % 8) * 8;
+        int c8 = (idx % HP) * 8;
         *reinterpret_cast<bf16x8*>(msg_out + (e0 + e) * H + c8) =
             lds8_silu(smem, L.z2 + (e * H_STRIDE + c8) * 2);
       }
     }
     {
-      f32x4 acc[4] = {};
-      mm_g<2, true>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      f32x4 acc[NT] = {};
+      mm_g<H / 32, true, NT>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                     opaque(w3), H, lane, acc);
       __bf16* z3 = reinterpret_cast<__bf16*>(smem + L.z3);
       float part[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
         float wv = biases[3 * H + c];
         float bb = biases[2 * H + c];
@@ -391,9 +413,9 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       sc[1] = dp;
     }
     if constexpr (FUSE_WG) __syncthreads();
-    {
-      // thread covers column c over 16 edges: dw3v partial + dz3 in place
-      int c = tid & 63;
+    for (int c = tid & 63; c < H; c += 64) {
+      // lane covers column(s) c over its wave's 16 edges: dw3v partial +
+      // dz3 in place
       int estart = (tid >> 6) * 16;
       float acc_w = 0.f, acc_b3 = 0.f;
       float wv = biases[3 * H + c];
@@ -418,10 +440,10 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       wg_acc<4, true>(smem, L.z3, H_STRIDE * 2, L.z2, H_STRIDE * 2, wave,
                       lane, wg3);
     } else {
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
         if (e >= nedge) continue;
-        int c8 = (idx % 8) * 8;
+        int c8 = (idx % HP) * 8;
         *reinterpret_cast<bf16x8*>(dz3_out + (e0 + e) * H + c8) =
             lds8(smem, L.z3 + (e * H_STRIDE + c8) * 2);
       }
@@ -429,15 +451,15 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
 
     // ---- dz2 = (dmsg_n[row] + dz3 @ W3) silu'(z2), overwrite z2 ----
     {
-      f32x4 acc[4] = {};
-      mm_g<2, false>(smem, L.z3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      f32x4 acc[NT] = {};
+      mm_g<H / 32, false, NT>(smem, L.z3 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w3t), H, lane, acc);
       if constexpr (FUSE_WG) __syncthreads();
       // z3 consumed: reuse its tile to stage dmsg_n[row] COALESCED
       // (the C-layout merge otherwise issues 16 scattered 2 B loads/lane)
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
-        int c8 = (idx % 8) * 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
+        int c8 = (idx % HP) * 8;
         bf16x8 v = {};
         if (e < nedge) v = g8(dmsg_n + (long)rws[e] * H + c8);
         *reinterpret_cast<bf16x8*>(smem + L.z3 + (e * H_STRIDE + c8) * 2) = v;
@@ -446,7 +468,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       __bf16* z2 = reinterpret_cast<__bf16*>(smem + L.z2);
       const __bf16* dmsg_t = reinterpret_cast<const __bf16*>(smem + L.z3);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -464,16 +486,15 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       wg_acc<4, true>(smem, L.z2, H_STRIDE * 2, L.z1, H_STRIDE * 2, wave,
                       lane, wg2);
     } else {
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
         if (e >= nedge) continue;
-        int c8 = (idx % 8) * 8;
+        int c8 = (idx % HP) * 8;
         *reinterpret_cast<bf16x8*>(dz2_out + (e0 + e) * H + c8) =
             lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
       }
     }
-    {
-      int c = tid & 63;
+    for (int c = tid & 63; c < H; c += 64) {
       int estart = (tid >> 6) * 16;
       const __bf16* z2 = reinterpret_cast<const __bf16*>(smem + L.z2);
       float acc_b = 0.f;
@@ -484,13 +505,13 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
 
     // ---- dz1 = (dz2 @ W2) silu'(z1), overwrite z1 ----
     {
-      f32x4 acc[4] = {};
-      mm_g<2, false>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
+      f32x4 acc[NT] = {};
+      mm_g<H / 32, false, NT>(smem, L.z2 + wave * 16 * H_STRIDE * 2, H_STRIDE * 2,
                      opaque(w2t), H, lane, acc);
       if constexpr (FUSE_WG) __syncthreads();
       __bf16* z1 = reinterpret_cast<__bf16*>(smem + L.z1);
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NT; ++nt) {
         int c = nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -502,16 +523,15 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     }
     if constexpr (FUSE_WG) __syncthreads();
     if constexpr (!FUSE_WG) {
-      for (int idx = lane; idx < 16 * 8; idx += 64) {
-        int e = wave * 16 + idx / 8;
+      for (int idx = lane; idx < 16 * HP; idx += 64) {
+        int e = wave * 16 + idx / HP;
         if (e >= nedge) continue;
-        int c8 = (idx % 8) * 8;
+        int c8 = (idx % HP) * 8;
         *reinterpret_cast<bf16x8*>(dz1_out + (e0 + e) * H + c8) =
             lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
       }
     }
-    {
-      int c = tid & 63;
+    for (int c = tid & 63; c < H; c += 64) {
       int estart = (tid >> 6) * 16;
       const __bf16* z1 = reinterpret_cast<const __bf16*>(smem + L.z1);
       float acc_b = 0.f;
@@ -528,19 +548,21 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       __syncthreads();
     }
 
-    // ---- dein = dz1 @ W1 in 3 register passes of 3 n-tiles ----
-    // pass outputs land in in_tile (ein no longer needed)
+    // ---- dein = dz1 @ W1 in register passes of <=3 n-tiles ----
+    // pass outputs land in in_tile (ein no longer needed). NTK 16-col
+    // tiles (9 at H=64) processed 3 per pass to bound live accumulators.
 #pragma unroll
-    for (int pass = 0; pass < 3; ++pass) {
+    for (int pass = 0; pass * 3 < NTK; ++pass) {
       const bf16* w1tp_ = opaque(w1tp);
       f32x4 acc[3] = {};
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
+      for (int kk = 0; kk < H / 32; ++kk) {
         int k = kk * 32 + (lane >> 4) * 8;
         bf16x8 a = lds8(smem, L.z1 + (wave * 16 + (lane & 15)) * H_STRIDE * 2
                                   + k * 2);
 #pragma unroll
         for (int nt = 0; nt < 3; ++nt) {
+          if (pass * 3 + nt >= NTK) break;
           int gc = (pass * 3 + nt) * 16 + (lane & 15);
           bf16x8 b = g8(w1tp_ + gc * H + k);
           acc[nt] =
@@ -550,6 +572,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
       __bf16* dein = reinterpret_cast<__bf16*>(smem + L.in_tile);
 #pragma unroll
       for (int nt = 0; nt < 3; ++nt) {
+        if (pass * 3 + nt >= NTK) break;
         int c = (pass * 3 + nt) * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -564,10 +587,10 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
     }
     if constexpr (FUSE_WG) __syncthreads();
     // dh_row / dh_col -> global (coalesced, wave-local rows)
-    for (int idx = lane; idx < 16 * 8; idx += 64) {
-      int e = wave * 16 + idx / 8;
+    for (int idx = lane; idx < 16 * HP; idx += 64) {
+      int e = wave * 16 + idx / HP;
       if (e >= nedge) continue;
-      int c8 = (idx % 8) * 8;
+      int c8 = (idx % HP) * 8;
       *reinterpret_cast<bf16x8*>(dhr_out + (e0 + e) * H + c8) =
           lds8(smem, L.in_tile + (e * K_STRIDE + c8) * 2);
       *reinterpret_cast<bf16x8*>(dhc_out + (e0 + e) * H + c8) =
@@ -614,7 +637,7 @@ __global__ __launch_bounds__(THREADS, FUSE_WG ? 2 : 3) void fused_edge_bwd(
 
 }  // namespace
 
-template <bool FUSE_WG>
+template <int H, bool FUSE_WG>
 static std::vector<torch::Tensor> fused_edge_backward_impl(
     torch::Tensor h, torch::Tensor coord, torch::Tensor eattr,
     torch::Tensor row, torch::Tensor col, torch::Tensor dmsg_n,
@@ -623,6 +646,9 @@ static std::vector<torch::Tensor> fused_edge_backward_impl(
     torch::Tensor w3v, bool normalize, double eps,
     std::vector<torch::Tensor> prepped) {
   long m = row.numel();
+  constexpr int K_IN = EDB<H>::K_IN;
+  constexpr int K_PAD = EDB<H>::K_PAD;
+  constexpr int K_OUT = EDB<H>::K_OUT;
   auto bopt = h.options();
   auto fopt = coord.options().dtype(torch::kFloat);
   long edge_rows = FUSE_WG ? 0 : m;  // per-edge buffers only when needed
@@ -645,7 +671,7 @@ static std::vector<torch::Tensor> fused_edge_backward_impl(
     return {ein, t1, msg, dz1, dz2, dz3, dhr, dhc, dcd, dw3v, gb};
   }
   auto stream = at::hip::getCurrentHIPStream();
-  constexpr Smem L = smem_layout();
+  constexpr Smem<H> L = smem_layout<H>();
   long tiles = (m + TILE - 1) / TILE;
   int blocks = (int)std::min<long>(tiles, 16384);
   auto hc = h.contiguous();
@@ -675,7 +701,7 @@ static std::vector<torch::Tensor> fused_edge_backward_impl(
     b3c = b3.contiguous().to(torch::kFloat);
     w3vc = w3v.contiguous().to(torch::kFloat);
   }
-  fused_edge_bwd<FUSE_WG><<<blocks, THREADS, L.total, stream>>>(
+  fused_edge_bwd<H, FUSE_WG><<<blocks, THREADS, L.total, stream>>>(
       reinterpret_cast<const bf16*>(hc.data_ptr()), cc.data_ptr<float>(),
       ec.data_ptr<float>(), row.contiguous().data_ptr<long>(),
       col.contiguous().data_ptr<long>(),
@@ -711,10 +737,20 @@ std::vector<torch::Tensor> fused_edge_backward(
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
     torch::Tensor w3v, bool normalize, double eps,
     std::vector<torch::Tensor> prepped) {
-  return fused_edge_backward_impl<false>(h, coord, eattr, row, col, dmsg_n,
-                                         dtrans_n, w1, b1, w2, b2, w3, b3,
-                                         w3v, normalize, eps,
-                                         std::move(prepped));
+  long hdim = h.size(1);
+  TORCH_CHECK(hdim == 32 || hdim == 64 || hdim == 128,
+              "fused edge backward supports hidden_nf in {32, 64, 128}");
+  if (hdim == 32)
+    return fused_edge_backward_impl<32, false>(
+        h, coord, eattr, row, col, dmsg_n, dtrans_n, w1, b1, w2, b2, w3,
+        b3, w3v, normalize, eps, std::move(prepped));
+  if (hdim == 128)
+    return fused_edge_backward_impl<128, false>(
+        h, coord, eattr, row, col, dmsg_n, dtrans_n, w1, b1, w2, b2, w3,
+        b3, w3v, normalize, eps, std::move(prepped));
+  return fused_edge_backward_impl<64, false>(
+      h, coord, eattr, row, col, dmsg_n, dtrans_n, w1, b1, w2, b2, w3, b3,
+      w3v, normalize, eps, std::move(prepped));
 }
 
 // wgrad-fused variant: returns {dhr, dhc, dcd, dw3v, gb, gw1, gw2, gw3} —
@@ -727,8 +763,9 @@ std::vector<torch::Tensor> fused_edge_backward_wg(
     torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
     torch::Tensor w3v, bool normalize, double eps,
     std::vector<torch::Tensor> prepped) {
-  return fused_edge_backward_impl<true>(h, coord, eattr, row, col, dmsg_n,
-                                        dtrans_n, w1, b1, w2, b2, w3, b3,
-                                        w3v, normalize, eps,
-                                        std::move(prepped));
+  TORCH_CHECK(h.size(1) == 64,
+              "the wgrad-fused edge backward is tuned for hidden_nf=64");
+  return fused_edge_backward_impl<64, true>(
+      h, coord, eattr, row, col, dmsg_n, dtrans_n, w1, b1, w2, b2, w3, b3,
+      w3v, normalize, eps, std::move(prepped));
 }

@@ -28,8 +28,12 @@ from typing import Callable, Dict, Tuple
 import torch
 import torch.nn.functional as F
 
-K_PAD = 160   # row padding of W1 for aligned 16-B B-fragments
-K_OUT = 144   # row count of the padded W1^T used by dgrad
+# Pad targets derive from the weight's input width (K_IN = w.size(1)):
+# K_PAD = next multiple of the 32-wide MFMA k-step (aligned 16-B
+# B-fragments), K_OUT = next multiple of 16 (dgrad n-tiles). H=64's
+# K_IN=131 reproduces the tuned constants 160/144; the same formulas
+# serve the H in {32, 128} edge-kernel instantiations and the virtual
+# block's K_IN = 129+C.
 
 
 def bf16c(t: torch.Tensor) -> torch.Tensor:
@@ -41,14 +45,16 @@ def f32c(t: torch.Tensor) -> torch.Tensor:
 
 
 def pad_kpad(t: torch.Tensor) -> torch.Tensor:
-    """[O, k] -> [O, K_PAD] bf16 (column zero-pad)."""
-    return F.pad(t.detach().bfloat16(), (0, K_PAD - t.size(1))).contiguous()
+    """[O, k] -> [O, roundup32(k)] bf16 (column zero-pad)."""
+    k_pad = (t.size(1) + 31) // 32 * 32
+    return F.pad(t.detach().bfloat16(), (0, k_pad - t.size(1))).contiguous()
 
 
 def tpad_kout(t: torch.Tensor) -> torch.Tensor:
-    """[O, k] -> [K_OUT, O] bf16 (transpose, row zero-pad)."""
+    """[O, k] -> [roundup16(k), O] bf16 (transpose, row zero-pad)."""
     tt = t.detach().bfloat16().t().contiguous()
-    return F.pad(tt, (0, 0, 0, K_OUT - tt.size(0))).contiguous()
+    k_out = (tt.size(0) + 15) // 16 * 16
+    return F.pad(tt, (0, 0, 0, k_out - tt.size(0))).contiguous()
 
 
 def t_bf16(t: torch.Tensor) -> torch.Tensor:
