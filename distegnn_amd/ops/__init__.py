@@ -492,8 +492,9 @@ class _FusedEdgeBlockFn(torch.autograd.Function):
                 gw2 = chunked_wgrad(dz2, t1).float()
                 gw3 = chunked_wgrad(dz3, msg).float()
             # bias grads accumulated in-kernel (block LDS + atomics):
-            # avoids three aten column-sum re-reads of [M, 64]
-            gb1, gb2, gb3 = gb[:64], gb[64:128], gb[128:]
+            # avoids three aten column-sum re-reads of [M, H]
+            hd = w1.size(0)
+            gb1, gb2, gb3 = gb[:hd], gb[hd:2 * hd], gb[2 * hd:]
             return (gh.to(h.dtype), gc, None, None, None, None, None, None,
                     gw1, gb1, gw2, gb2, gw3, gb3, gw3v, None, None)
         # fallback: recompute the eager composition under autograd

@@ -413,6 +413,7 @@ void fused_edge_bwd(
       sc[1] = dp;
     }
     if constexpr (FUSE_WG) __syncthreads();
+#pragma unroll
     for (int c = tid & 63; c < H; c += 64) {
       // lane covers column(s) c over its wave's 16 edges: dw3v partial +
       // dz3 in place
@@ -494,6 +495,7 @@ void fused_edge_bwd(
             lds8(smem, L.z2 + (e * H_STRIDE + c8) * 2);
       }
     }
+#pragma unroll
     for (int c = tid & 63; c < H; c += 64) {
       int estart = (tid >> 6) * 16;
       const __bf16* z2 = reinterpret_cast<const __bf16*>(smem + L.z2);
@@ -531,6 +533,7 @@ void fused_edge_bwd(
             lds8(smem, L.z1 + (e * H_STRIDE + c8) * 2);
       }
     }
+#pragma unroll
     for (int c = tid & 63; c < H; c += 64) {
       int estart = (tid >> 6) * 16;
       const __bf16* z1 = reinterpret_cast<const __bf16*>(smem + L.z1);
