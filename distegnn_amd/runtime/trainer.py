@@ -255,7 +255,15 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
                             grad_bucket.graph_sync()
                         else:
                             grad_bucket.sync()      # SUM == ref avg * ws
-                    if ((world_size > 1 or dataset_name == "LargeFluid")
+                    # reference clip rule (utils/train.py:153) tests
+                    # dataset_name == 'LargeFluid', but the reference's own
+                    # headline config names the dataset 'Fluid113K'
+                    # (config/largefluid_distegnn.yaml:14) — so its
+                    # single-GPU fluid runs never clip and can diverge at
+                    # lr 5e-4 (reproduced: NaN by epoch 6). Deliberate
+                    # deviation: accept both names.
+                    if ((world_size > 1
+                         or dataset_name in ("LargeFluid", "Fluid113K"))
                             and model_name == "FastEGNN"):
                         nn.utils.clip_grad_norm_(model.parameters(),
                                                  max_norm=0.3)
@@ -420,8 +428,8 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                                 grad_bucket.graph_sync()
                             else:
                                 grad_bucket.sync()
-                        if ((world_size > 1 or
-                             config.data.dataset_name == "LargeFluid")
+                        if ((world_size > 1 or config.data.dataset_name
+                             in ("LargeFluid", "Fluid113K"))
                                 and model_name == "FastEGNN"):
                             nn.utils.clip_grad_norm_(model.parameters(),
                                                      max_norm=0.3)
