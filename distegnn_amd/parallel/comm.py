@@ -253,6 +253,24 @@ class CapturedAllReduce:
             if t.is_cuda and self._key(t, op) not in self._entries:
                 self._build(t, op)
 
+    def rebuild(self):
+        """Drop + rebuild every captured all-reduce graph. Call after any
+        device-wide sync between replays (eval epochs, checkpoint D2H):
+        a hipDeviceSynchronize garbles existing graph execs on this stack
+        (runtime/graphs.GraphedStep.invalidate). Rebuild EAGERLY (not
+        lazily) so the builds' own pre-capture syncs land before any step
+        graph recaptures."""
+        if not self._entries:
+            return
+        specs = [(buf, key[2]) for key, (g, buf) in self._entries.items()]
+        self._entries = {}
+        for buf, op_str in specs:
+            for op in (dist.ReduceOp.SUM, dist.ReduceOp.MAX,
+                       dist.ReduceOp.MIN, dist.ReduceOp.AVG):
+                if str(op) == op_str:
+                    self._build(buf, op)
+                    break
+
     def __call__(self, t: torch.Tensor, op=None) -> torch.Tensor:
         op = op if op is not None else dist.ReduceOp.SUM
         if not is_distributed():
@@ -336,6 +354,13 @@ class GradBucket:
         barrier) must not land inside the replay window."""
         if is_distributed() and torch.cuda.is_available():
             self.graph_sync(_build_only=True)
+
+    def rebuild_graph_sync(self):
+        """Drop + eagerly rebuild the captured grad-sync graph after a
+        device-wide sync between replays (see CapturedAllReduce.rebuild)."""
+        if getattr(self, "_sync_graph", None) is not None:
+            self._sync_graph = None
+            self.prebuild_graph_sync()
 
     @torch.no_grad()
     def graph_sync(self, _build_only: bool = False):

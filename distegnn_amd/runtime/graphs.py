@@ -110,11 +110,52 @@ class GraphedStep:
             return fn()
         if self._side is None:
             self._side = torch.cuda.Stream()
+            # Dedicated allocator pool for eager work between replays.
+            # The side stream alone is NOT sufficient isolation: a large
+            # eager region (an fp32 eval epoch at 113K nodes allocates
+            # hundreds of MB) was bisected to corrupt replayed training
+            # state on this stack even on the side stream — the fourth
+            # replay-window hazard. Allocations drawn from a private
+            # MemPool can never alias blocks the captured graphs
+            # reference. (Fallback: side stream only, as in round 1.)
+            try:
+                self._eager_pool = torch.cuda.MemPool()
+            except Exception:
+                self._eager_pool = None
         self._side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(self._side):
-            out = fn()
+            if self._eager_pool is not None:
+                with torch.cuda.memory.use_mem_pool(self._eager_pool):
+                    out = fn()
+            else:
+                out = fn()
         torch.cuda.current_stream().wait_stream(self._side)
         return out
+
+    def invalidate(self, reason: str = ""):
+        """Drop every captured graph; each shape recaptures on its next
+        occurrence (no extra eager warmups — out_meta is kept).
+
+        MUST be called after any operation that device-synchronizes
+        between replays. On this ROCm stack a hipDeviceSynchronize
+        garbles existing graph execs: a bare torch.cuda.synchronize()
+        (or a pageable D2H read, e.g. torch.save / .cpu() on params)
+        between replays NaNs the training trajectory about one epoch
+        later even though the very next epoch of replays is still
+        bit-exact (bisected via DISTEGNN_DBG_EVAL_MODE=sync/d2h;
+        profiles/README.md replay-window hazard #3). Eval epochs and
+        checkpoint saves therefore invalidate + recapture."""
+        n = 0
+        for e in self.entries.values():
+            if e.graph is not None:
+                e.graph = None
+                e.static = None
+                e.outputs = None
+                n += 1
+        if n and self.verbose:
+            print(f"[graphs] invalidated {n} captured graph(s)"
+                  f"{': ' + reason if reason else ''}")
+        return n
 
     def _fallback(self, batch):
         if self.fallback_ctx is None:

@@ -235,7 +235,9 @@ def train_single_epoch(rank, model, model_name, loader, optimizer, scheduler,
             loss_accum.add_(loss_loc.detach(), alpha=batch_size)
             counter.add_(batch_size)
 
-            if _is_fast_model(model_name) and virtual_node_loc is not None:
+            if (_is_fast_model(model_name) and virtual_node_loc is not None
+                    and not (not backprop and os.environ.get(
+                        "DISTEGNN_DBG_EVAL_MODE") == "nommd")):
                 vloc = virtual_node_loc.permute(0, 2, 1).float()  # [B,C,3]
                 lm = mmd_loss(vloc, data.target, data.batch, data.ptr,
                               data.counts, train_config.mmd.sigma,
@@ -484,7 +486,13 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
         if rank == 0:
             log_dict["loss_train"].append(loss_train)
 
-        if epoch_index % log_config.test_interval == 0:
+        # debug-only bisection knob for the eval-epoch block (used to pin
+        # down which part of the block perturbs captured-graph replays):
+        # full | skipall | skip (ckpt only) | nockpt (eval only) |
+        # loader (H2D iteration only, no forward) | nommd (eval w/o MMD)
+        _dbg_eval = os.environ.get("DISTEGNN_DBG_EVAL_MODE", "full")
+        if (epoch_index % log_config.test_interval == 0
+                and _dbg_eval != "skipall"):
             # the reference evaluates in fp32 (no autocast in
             # utils/train.py's no-grad epochs): eval/checkpoint selection
             # stays fp32 unless train.bf16_eval is set explicitly
@@ -506,7 +514,40 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                         autocast_dtype=eval_dtype, progress=progress,
                         epoch_reduce=epoch_reduce)
 
-            if graphed_step is not None:
+            if _dbg_eval == "skip":
+                loss_valid = loss_test = float(loss_train)
+            elif _dbg_eval in ("sync", "gc", "save", "osd", "d2h", "sleep"):
+                # micro-triggers: which single operation garbles replays?
+                if _dbg_eval == "sync":
+                    torch.cuda.synchronize()
+                elif _dbg_eval == "gc":
+                    import gc as _gc
+                    _gc.collect()
+                elif _dbg_eval == "save":
+                    torch.save(state_dict_for_save(model, world_size),
+                               "/tmp/_dbg_ckpt.pth")
+                elif _dbg_eval == "osd":
+                    optimizer.state_dict()
+                elif _dbg_eval == "d2h":
+                    for _p in model.parameters():
+                        _p.detach().cpu()
+                elif _dbg_eval == "sleep":
+                    time.sleep(2.0)
+                loss_valid = loss_test = float(loss_train)
+            elif _dbg_eval == "loader":
+                def _iterate(loader):
+                    for d in loader:
+                        d.to(device)
+                    return float(loss_train)
+                if graphed_step is not None:
+                    loss_valid = graphed_step.run_eager(
+                        lambda: _iterate(loader_valid))
+                    loss_test = graphed_step.run_eager(
+                        lambda: _iterate(loader_test))
+                else:
+                    loss_valid = _iterate(loader_valid)
+                    loss_test = _iterate(loader_test)
+            elif graphed_step is not None:
                 # eval allocations must not alias captured-graph pools
                 loss_valid = graphed_step.run_eager(
                     lambda: _eval(loader_valid, "valid"))
@@ -516,7 +557,10 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                 loss_valid = _eval(loader_valid, "valid")
                 loss_test = _eval(loader_test, "test")
 
-            if rank == 0:
+            if rank == 0 and _dbg_eval in ("nockpt", "nommd", "loader", "sync", "gc", "save", "osd", "d2h", "sleep"):
+                print(f"*** Best Valid Loss: {loss_valid:.5f}"
+                      f" | (dbg {_dbg_eval}: checkpoint skipped)")
+            if rank == 0 and _dbg_eval not in ("nockpt", "nommd", "loader", "sync", "gc", "save", "osd", "d2h", "sleep"):
                 log_dict["epochs"].append(epoch_index)
                 log_dict["loss"].append(loss_test)
                 state = {
@@ -546,6 +590,20 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
             if rank == 0 and wb is not None:
                 wb.log({"loss_train": loss_train, "loss_valid": loss_valid,
                         "loss_test": loss_test, "epoch": epoch_index})
+
+            # The eval epoch device-synchronized (eval forwards, checkpoint
+            # D2H in torch.save): a device-wide sync between replays garbles
+            # existing hipGraph execs on this stack (bisected — a bare
+            # torch.cuda.synchronize() here NaNs the trajectory one epoch
+            # later; see GraphedStep.invalidate). Drop and recapture
+            # EVERYTHING before the next replay, comm graphs first so their
+            # build-time syncs precede the step recaptures.
+            if graphed_step is not None and graphed_step.enabled:
+                graphed_step.invalidate("post-eval device sync")
+                if grad_bucket is not None:
+                    grad_bucket.rebuild_graph_sync()
+                if epoch_reduce is not None:
+                    epoch_reduce.rebuild()
             if rank == 0 and (epoch_index - best_log_dict["epoch_index"]
                               >= train_config.early_stop):
                 best_log_dict["early_stop"] = epoch_index
