@@ -288,6 +288,11 @@ def main():
         run_capture_integrity_gate(graphed, _gate_step, n_check,
                                    list(model.parameters()), optimizer,
                                    rank=rank)
+        # the gate's final D2H device-syncs, garbling prebuilt graph execs
+        # (GraphedStep.invalidate): the gate dropped its step captures;
+        # rebuild the captured grad sync too. Warmup below recaptures.
+        if graphed.enabled and ws_eff > 1 and grad_bucket is not None:
+            grad_bucket.rebuild_graph_sync()
 
     mse = None
     # a shape's graph is captured at its (warmup_occurrences+1)-th

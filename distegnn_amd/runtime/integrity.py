@@ -97,6 +97,12 @@ def run_capture_integrity_gate(
         if p.grad is None:
             p.grad = torch.zeros_like(p)
     snap = _snapshot_state(params, optimizer)
+    # the gate reseeds and consumes RNG; restore the global streams at the
+    # end so a gated run follows the SAME training trajectory as an
+    # ungated one (trajectory parity is itself a test invariant)
+    rng_cpu = torch.get_rng_state()
+    rng_cuda = (torch.cuda.get_rng_state(device)
+                if device.type == "cuda" else None)
 
     # Phase 1: eager reference trajectory (no graphs exist yet — default
     # stream/communicator are still safe here)
@@ -151,6 +157,19 @@ def run_capture_integrity_gate(
 
     # final restore happens between replays — side stream discipline
     graphed.run_eager(lambda: _restore_state(params, optimizer, snap))
+
+    torch.set_rng_state(rng_cpu)
+    if rng_cuda is not None:
+        torch.cuda.set_rng_state(rng_cuda, device)
+
+    # the gate itself just did a pageable D2H (.cpu() of the loss buffer):
+    # a device-wide sync garbles existing graph execs on this stack
+    # (GraphedStep.invalidate docstring) — drop the gate's captures so
+    # production replays use fresh ones. Callers holding prebuilt comm
+    # graphs (GradBucket.graph_sync, CapturedAllReduce) must rebuild them
+    # after the gate for the same reason.
+    if hasattr(graphed, "invalidate"):
+        graphed.invalidate("post-gate D2H sync")
 
     if not ok:
         graphed.enabled = False

@@ -449,6 +449,14 @@ def train(rank, model, model_name, optimizer, scheduler, loader_train,
                 n_steps=(graphed_step.warmup + 2) * len(gate_batches),
                 params=list(model.parameters()), optimizer=optimizer,
                 rank=rank)
+            # the gate's final D2H garbles prebuilt graph execs
+            # (GraphedStep.invalidate) — the gate drops its own step
+            # captures; rebuild the comm graphs it cannot see
+            if graphed_step.enabled:
+                if grad_bucket is not None:
+                    grad_bucket.rebuild_graph_sync()
+                if epoch_reduce is not None:
+                    epoch_reduce.rebuild()
 
     # DISTEGNN_TORCH_PROFILE=<dir>: trace the FIRST epoch after warmup with
     # torch.profiler (chrome trace per rank). rocprofv3 stays the primary

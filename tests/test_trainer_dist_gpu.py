@@ -110,3 +110,50 @@ def test_graphed_multigpu_trainer_on_one_rank_group(tmp_path, capsys):
     assert hist["loss_train"] and all(math.isfinite(v)
                                       for v in hist["loss_train"])
     assert math.isfinite(best["loss_valid"])
+
+
+@pytest.mark.timeout(600)
+def test_graphed_eval_epochs_match_eager(tmp_path):
+    """Regression for the eval-epoch replay hazard: eval forwards and
+    checkpoint saves device-synchronize between replays, which garbles
+    existing hipGraph execs on this stack (a bare torch.cuda.synchronize()
+    there NaNs training one epoch later — bisected via
+    DISTEGNN_DBG_EVAL_MODE). The trainer must invalidate + recapture after
+    every eval epoch; with that, a graphed run with interleaved evals
+    follows the eager trajectory."""
+    _init_one_rank_group(tmp_path)
+    device = torch.device("cuda:0")
+
+    def _run(hip_graphs):
+        fix_seed(43)
+        cfg = _config(tmp_path)
+        cfg.train.epochs = 6
+        cfg.train.hip_graphs = hip_graphs
+        # gate off: it consumes loader batches pre-training, which would
+        # shift the graphed arm's epoch-1 batch order vs the eager arm
+        # (gate coverage lives in the test above)
+        cfg.train.graph_integrity_check = "off"
+        cfg.log.test_interval = 2          # evals+checkpoints at 2, 4, 6
+        cfg.log.exp_name = f"evalpar_{hip_graphs}"
+        model = FastEGNN(node_feat_nf=2, node_attr_nf=0, edge_attr_nf=2,
+                         hidden_nf=32, virtual_channels=3, world_size=2,
+                         n_layers=2, normalize=False).to(device)
+        grad_bucket = GradBucket(model)
+        optimizer = torch.optim.Adam(model.parameters(), lr=5e-4,
+                                     weight_decay=1e-12)
+        train_ds = DatasetWrapper(_partition_samples(6))
+        eval_ds = DatasetWrapper(_partition_samples(2, seed=12))
+        lt, lv, ltst = make_loaders(train_ds, eval_ds, eval_ds, 1, seed=43)
+        _, hist = trainer.train(
+            0, model, "FastEGNN", optimizer, None, lt, lv, ltst,
+            cfg.train, cfg.log, cfg, start_epoch=0, device=device,
+            grad_bucket=grad_bucket, autocast_dtype=torch.bfloat16,
+            progress=False)
+        return hist["loss_train"]
+
+    eager = _run("off")
+    graphed = _run("on")
+    assert all(math.isfinite(v) for v in graphed), graphed
+    assert len(eager) == len(graphed) == 6
+    for e, g in zip(eager, graphed):
+        assert abs(e - g) <= 5e-3 * max(abs(e), 1e-8), (eager, graphed)
