@@ -121,10 +121,14 @@ def test_graphed_eval_epochs_match_eager(tmp_path):
     DISTEGNN_DBG_EVAL_MODE). The trainer must invalidate + recapture after
     every eval epoch; with that, a graphed run with interleaved evals
     follows the eager trajectory."""
-    _init_one_rank_group(tmp_path)
     device = torch.device("cuda:0")
 
     def _run(hip_graphs):
+        # trainer.train tears the process group down at the end of every
+        # world_size>1 run — re-init per arm (fresh file store each time)
+        d = tmp_path / f"pg_{hip_graphs}"
+        d.mkdir(exist_ok=True)
+        _init_one_rank_group(d)
         fix_seed(43)
         cfg = _config(tmp_path)
         cfg.train.epochs = 6
