@@ -149,7 +149,19 @@ __global__ __launch_bounds__(THREADS, H <= 64 ? 2 : 1) void fused_edge_fwd(
 
   // Wave-per-subtile execution: every phase reads/writes only its own
   // wave's 16 rows, so the tile loop carries NO barriers.
-  for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
+  //
+  // XCD-aware tile remap: the dispatcher places workgroup b on XCD b%8
+  // (each XCD has a private 4 MiB L2). In identity order neighboring
+  // 64-edge tiles — whose src-sorted row gathers share h/x rows
+  // (~14.6 edges/node) — land on DIFFERENT XCDs, so every XCD streams
+  // the whole 15 MB feature table. The bijective remap below gives each
+  // XCD a CONTIGUOUS tile range (~1/8 of the rows ≈ 1.9 MB, L2-fits).
+  const long ntile = (m + TILE - 1) / TILE;
+  const long tq = ntile >> 3, tr = ntile & 7;
+  for (long vt = blockIdx.x; vt < ntile; vt += gridDim.x) {
+    const long xcd = vt & 7, ti = vt >> 3;
+    const long tile =
+        (xcd < tr ? xcd * (tq + 1) : tr * (tq + 1) + (xcd - tr) * tq) + ti;
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
     int* rws = reinterpret_cast<int*>(smem + L.rows);

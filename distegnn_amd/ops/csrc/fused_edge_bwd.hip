@@ -255,7 +255,14 @@ void fused_edge_bwd(
   // SQ_WAIT_ANY dominated by the 14 sync-separated phases). The FUSE_WG
   // variant's wgrad contraction runs over the whole 64-edge tile, so it
   // keeps the original barrier schedule.
-  for (long tile = blockIdx.x; tile * TILE < m; tile += gridDim.x) {
+  // XCD-aware tile remap (same as fused_edge_fwd): contiguous tile
+  // ranges per XCD keep each XCD's row gathers inside its 4 MiB L2.
+  const long ntile = (m + TILE - 1) / TILE;
+  const long tq = ntile >> 3, tr = ntile & 7;
+  for (long vt = blockIdx.x; vt < ntile; vt += gridDim.x) {
+    const long xcd = vt & 7, ti = vt >> 3;
+    const long tile =
+        (xcd < tr ? xcd * (tq + 1) : tr * (tq + 1) + (xcd - tr) * tq) + ti;
     long e0 = tile * TILE;
     int nedge = (int)((m - e0 < (long)TILE) ? (m - e0) : (long)TILE);
     if constexpr (FUSE_WG) __syncthreads();
