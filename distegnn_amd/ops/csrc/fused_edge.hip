@@ -103,7 +103,8 @@ __device__ __forceinline__ void mm_a_lds(const char* smem, int a_off,
                                          int a_stride,
                                          const bf16* __restrict__ w, int wk,
                                          int lane, f32x4 (&acc)[NT]) {
-#pragma unroll
+__builtin_amdgcn_s_setprio(1);
+  #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
     bf16x8 a = lds8(smem, a_off + (lane & 15) * a_stride + k * 2);
@@ -113,6 +114,7 @@ __device__ __forceinline__ void mm_a_lds(const char* smem, int a_off,
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
+  __builtin_amdgcn_s_setprio(0);
 }
 
 template <int H>

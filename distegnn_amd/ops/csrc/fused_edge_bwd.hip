@@ -145,7 +145,8 @@ __device__ __forceinline__ void wg_acc(const char* smem, int a_base,
                                        int a_stride, int b_base,
                                        int b_stride, int wave, int lane,
                                        f32x4 (&acc)[NJ]) {
-#pragma unroll
+__builtin_amdgcn_s_setprio(1);
+  #pragma unroll
   for (int kk = 0; kk < 2; ++kk) {
     int k0 = kk * 32 + (lane >> 4) * 8;
     bf16x8 a = lds8_t(smem, a_base, a_stride, wave * 16 + (lane & 15), k0);
@@ -160,6 +161,7 @@ __device__ __forceinline__ void wg_acc(const char* smem, int a_base,
           __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[j], 0, 0, 0);
     }
   }
+  __builtin_amdgcn_s_setprio(0);
 }
 
 // A from LDS (optionally through silu), B from global [64][wk] k-contig.
@@ -168,7 +170,8 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
                                      int a_stride,
                                      const bf16* __restrict__ w, int wk,
                                      int lane, f32x4 (&acc)[NT]) {
-#pragma unroll
+__builtin_amdgcn_s_setprio(1);
+  #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
     bf16x8 a = SILU_A ? lds8_silu(smem, a_off + (lane & 15) * a_stride + k * 2)
@@ -179,6 +182,7 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
+  __builtin_amdgcn_s_setprio(0);
 }
 
 // FUSE_WG: accumulate the three weight gradients (dW1 = dz1^T ein,
