@@ -71,6 +71,7 @@ template <int KSTEPS, bool SILU_A>
 __device__ __forceinline__ void mm_g(const char* smem, int a_off,
                                      int a_stride, const bf16* w, int wk,
                                      int lane, f32x4 (&acc)[4]) {
+  __builtin_amdgcn_s_setprio(1);
 #pragma unroll
   for (int kk = 0; kk < KSTEPS; ++kk) {
     int k = kk * 32 + (lane >> 4) * 8;
@@ -82,6 +83,7 @@ __device__ __forceinline__ void mm_g(const char* smem, int a_off,
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
     }
   }
+  __builtin_amdgcn_s_setprio(0);
 }
 
 struct VSmem {
