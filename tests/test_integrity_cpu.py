@@ -139,6 +139,39 @@ def test_graphedstep_key_distinguishes_optional_fields():
     assert k_chunk3 == GraphedStep._key(mk(True, 3))
 
 
+def test_graphedstep_invalidate_drops_graphs_keeps_meta():
+    """invalidate() (the device-sync replay-hazard mitigation) must drop
+    every captured graph but keep shape statistics/out_meta, so each shape
+    recaptures on its next occurrence without extra eager warmups."""
+    from distegnn_amd.runtime.graphs import GraphedStep, _ShapeEntry
+
+    g = GraphedStep(lambda b: (torch.zeros(()),), [], enabled=False)
+    e1, e2, e3 = _ShapeEntry(), _ShapeEntry(), _ShapeEntry()
+    e1.graph = object()
+    e1.static = {"x": torch.zeros(1)}
+    e1.outputs = (torch.zeros(()),)
+    e1.seen = 5
+    e1.out_meta = [((1,), torch.float32, "cpu")]
+    e2.seen = 1                       # still warming up — untouched
+    e3.disabled = True                # capture failed earlier — untouched
+    g.entries = {"a": e1, "b": e2, "c": e3}
+
+    assert g.invalidate("test") == 1
+    assert e1.graph is None and e1.static is None and e1.outputs is None
+    assert e1.seen == 5 and e1.out_meta is not None   # recapture-ready
+    assert e2.seen == 1 and not e2.disabled
+    assert e3.disabled
+    assert g.invalidate() == 0        # idempotent
+
+
+def test_captured_allreduce_rebuild_noop_without_entries():
+    from distegnn_amd.parallel.comm import CapturedAllReduce
+
+    red = CapturedAllReduce()
+    red.rebuild()                     # no entries, no dist: must not throw
+    assert red._entries == {}
+
+
 # ---------------------------------------------------------------------------
 # 2-rank gloo: gate verdict coherence + CapturedAllReduce eager fallback
 
