@@ -265,11 +265,16 @@ class CapturedAllReduce:
         specs = [(buf, key[2]) for key, (g, buf) in self._entries.items()]
         self._entries = {}
         for buf, op_str in specs:
-            for op in (dist.ReduceOp.SUM, dist.ReduceOp.MAX,
-                       dist.ReduceOp.MIN, dist.ReduceOp.AVG):
-                if str(op) == op_str:
-                    self._build(buf, op)
-                    break
+            op = next((o for o in (dist.ReduceOp.SUM, dist.ReduceOp.MAX,
+                                   dist.ReduceOp.MIN, dist.ReduceOp.AVG,
+                                   dist.ReduceOp.PRODUCT)
+                       if str(o) == op_str), None)
+            if op is None:
+                # dropping an entry silently would desync the per-rank
+                # collective schedule on the next replay
+                raise RuntimeError(
+                    f"CapturedAllReduce.rebuild: unknown op {op_str!r}")
+            self._build(buf, op)
 
     def __call__(self, t: torch.Tensor, op=None) -> torch.Tensor:
         op = op if op is not None else dist.ReduceOp.SUM
