@@ -22,8 +22,10 @@ def build_model(name, device):
     from distegnn_amd.models.fastrf import FastRF
     from distegnn_amd.models.fastschnet import FastSchNet
     from distegnn_amd.models.fasttfn import FastTFN
-    from distegnn_amd.models.baselines import EGNN, RF_vel, Linear_dynamics
+    from distegnn_amd.models.baselines import (EGNN, RF_vel,
+                                                Linear_dynamics, EGHN)
     from distegnn_amd.models.schnet import SchNet
+    from distegnn_amd.models.tfn import OurDynamics
 
     common = dict(hidden_nf=64, virtual_channels=3, n_layers=4)
     if name == "FastEGNN":
@@ -45,6 +47,12 @@ def build_model(name, device):
         m = RF_vel(hidden_nf=64, edge_attr_nf=2, n_layers=4)
     elif name == "SchNet":
         m = SchNet(hidden_channels=64, num_filters=64, num_interactions=4)
+    elif name == "TFN":
+        m = OurDynamics(nf=32, n_layers=4, model="tfn", num_degrees=2,
+                        div=1)
+    elif name == "EGHN":
+        m = EGHN(in_node_nf=2, in_edge_nf=2, hidden_nf=64, n_cluster=4,
+                 layer_per_block=2, layer_pooling=2)
     elif name == "Linear":
         m = Linear_dynamics()
     else:
@@ -55,6 +63,14 @@ def build_model(name, device):
 def run_step(name, model, batch, device):
     from distegnn_amd.runtime.trainer import model_forward
 
+    if name == "EGHN":
+        n_node = batch.num_nodes // batch.num_graphs
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            enabled=device.type == "cuda"):
+            pred, _, _ = model(batch.pos, batch.x, batch.edge_index,
+                               batch.edge_attr, batch.edge_index,
+                               batch.edge_attr, n_node, v=batch.vel)
+        return torch.nn.functional.mse_loss(pred.float(), batch.target)
     key = {"RF_vel": "RF", "Linear": "Linear"}.get(name, name)
     with torch.autocast("cuda", dtype=torch.bfloat16,
                         enabled=device.type == "cuda"):
@@ -70,7 +86,7 @@ def main():
     ap.add_argument("--graphs-per-batch", type=int, default=4)
     ap.add_argument("--models", type=str,
                     default="FastEGNN,FastRF,FastSchNet,FastTFN,EGNN,"
-                            "RF_vel,SchNet,Linear")
+                            "RF_vel,SchNet,TFN,EGHN,Linear")
     args = ap.parse_args()
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
 
